@@ -1,0 +1,148 @@
+"""Device-plugin manager tests.
+
+Analog of nvidia_gpu_manager_test.go:100-150: a fake backend with an
+8-GPU topology fixture and a degenerate no-topology fixture; asserts the
+exact advertised resource-name tree and the allocation outputs (device
+nodes + ROCR_VISIBLE_DEVICES instead of NVIDIA_VISIBLE_DEVICES / REST
+daemon CLI parsing).
+"""
+
+import pytest
+
+from kubegpu_amd.api.types import ContainerInfo, NodeInfo, PodInfo
+from kubegpu_amd.deviceplugin import create_device_plugin
+from kubegpu_amd.discovery import (
+    CrashingBackend,
+    FakeBackend,
+    GpusInfo,
+    fixtures,
+)
+from kubegpu_amd.plugintypes import RESOURCE_GPU
+
+VRAM = fixtures.MI355X_VRAM_BYTES
+
+
+def _mgr(fix):
+    return create_device_plugin(FakeBackend(fix))
+
+
+def test_full_hive_advertisement(fixture_8x):
+    mgr = _mgr(fixture_8x)
+    mgr.start()
+    ni = NodeInfo(name="n")
+    mgr.update_node_info(ni)
+    assert ni.kube_cap[RESOURCE_GPU] == 8
+    assert ni.allocatable[RESOURCE_GPU] == 8
+    # full mesh -> single gpugrp0/gpugrp1 group
+    for i in range(8):
+        uuid = f"GPU-mi355x-{i:02d}"
+        cards = f"resource/group/gpugrp1/0/gpugrp0/0/gpu/{uuid}/cards"
+        mem = f"resource/group/gpugrp1/0/gpugrp0/0/gpu/{uuid}/memory"
+        assert ni.allocatable[cards] == 1
+        assert ni.allocatable[mem] == VRAM
+    # 8 cards + 8 memory + flat count
+    assert len(ni.allocatable) == 17
+
+
+def test_two_hive_grouping(fixture_2hive):
+    mgr = _mgr(fixture_2hive)
+    mgr.start()
+    ni = NodeInfo(name="n")
+    mgr.update_node_info(ni)
+    # GPUs 0-3 in gpugrp0/0, 4-7 in gpugrp0/1; numa splits match hives so
+    # gpugrp1 has two groups as well
+    assert "resource/group/gpugrp1/0/gpugrp0/0/gpu/GPU-mi355x-00/cards" in ni.allocatable
+    assert "resource/group/gpugrp1/1/gpugrp0/1/gpu/GPU-mi355x-04/cards" in ni.allocatable
+
+
+def test_degenerate_no_topology(fixture_no_xgmi):
+    """No xGMI: each GPU its own gpugrp0, one shared-NUMA gpugrp1 (analog
+    of the "Topology":null K80 fixture, nvidia_gpu_manager_test.go:140-144)."""
+    mgr = _mgr(fixture_no_xgmi)
+    mgr.start()
+    ni = NodeInfo(name="n")
+    mgr.update_node_info(ni)
+    for i in range(4):
+        uuid = f"GPU-mi355x-{i:02d}"
+        assert (
+            f"resource/group/gpugrp1/0/gpugrp0/{i}/gpu/{uuid}/cards" in ni.allocatable
+        )
+
+
+def test_allocate_devices_and_env(fixture_8x):
+    mgr = _mgr(fixture_8x)
+    mgr.start()
+    cont = ContainerInfo(
+        allocate_from={
+            "resource/group/gpugrp1/0/gpugrp0/0/gpu/2/cards":
+                "resource/group/gpugrp1/0/gpugrp0/0/gpu/GPU-mi355x-02/cards",
+            "resource/group/gpugrp1/0/gpugrp0/0/gpu/3/cards":
+                "resource/group/gpugrp1/0/gpugrp0/0/gpu/GPU-mi355x-05/cards",
+        }
+    )
+    mounts, devices, envs = mgr.allocate(PodInfo(name="p"), cont)
+    assert mounts == []
+    assert "/dev/kfd" in devices
+    assert "/dev/dri/renderD130" in devices  # index 2
+    assert "/dev/dri/renderD133" in devices  # index 5
+    assert envs["ROCR_VISIBLE_DEVICES"] == "GPU-mi355x-02,GPU-mi355x-05"
+    assert mgr.gpus["GPU-mi355x-02"].in_use
+
+
+def test_allocate_unknown_gpu_raises(fixture_8x):
+    mgr = _mgr(fixture_8x)
+    mgr.start()
+    cont = ContainerInfo(
+        allocate_from={
+            "r": "resource/group/gpugrp1/0/gpugrp0/0/gpu/GPU-nonexistent/cards"
+        }
+    )
+    with pytest.raises(KeyError):
+        mgr.allocate(PodInfo(name="p"), cont)
+
+
+def test_crash_containment():
+    """Discovery failure must not fail start; node advertises 0 GPUs
+    (cf. nvidia_gpu_manager.go:185-188,193-197)."""
+    mgr = create_device_plugin(CrashingBackend())
+    mgr.start()  # no raise
+    ni = NodeInfo(name="n")
+    mgr.update_node_info(ni)
+    assert ni.kube_cap[RESOURCE_GPU] == 0
+    assert len(ni.allocatable) == 1
+
+
+def test_mark_sweep_rediscovery(fixture_8x):
+    """A GPU vanishing between updates is swept; in_use survives for the
+    rest (nvidia_gpu_manager.go:132-155)."""
+    backend = FakeBackend(fixture_8x)
+    mgr = create_device_plugin(backend)
+    mgr.start()
+    assert len(mgr.gpus) == 8
+    mgr.gpus["GPU-mi355x-01"].in_use = True
+
+    smaller = GpusInfo.from_json(fixture_8x.to_json())
+    smaller.devices = [d for d in smaller.devices if d.index != 7]
+    backend.set_info(smaller)
+    mgr.update_gpu_info(force=True)
+    assert len(mgr.gpus) == 7
+    assert "GPU-mi355x-07" not in mgr.gpus
+    assert mgr.gpus["GPU-mi355x-01"].in_use  # preserved
+
+
+def test_discovery_cache(fixture_8x):
+    """Within the 5-minute window, the backend is not re-queried."""
+    calls = {"n": 0}
+
+    class CountingBackend(FakeBackend):
+        def get_gpu_info(self):
+            calls["n"] += 1
+            return super().get_gpu_info()
+
+    mgr = create_device_plugin(CountingBackend(fixture_8x))
+    mgr.start()
+    mgr.update_gpu_info()
+    mgr.update_gpu_info()
+    assert calls["n"] == 1
+    mgr.update_gpu_info(force=True)
+    assert calls["n"] == 2

@@ -1,0 +1,133 @@
+"""Request-translation tests (cf. gpuschedulerplugin/gpu.go:16-127)."""
+
+import pytest
+
+from kubegpu_amd.api.types import ContainerInfo, NodeInfo, PodInfo
+from kubegpu_amd.plugintypes import RESOURCE_GPU
+from kubegpu_amd.scheduler import (
+    GPU_TOPOLOGY_GENERATION,
+    SchedulingError,
+    TWO_LEVEL_TEMPLATE,
+    NodeTreeCache,
+    set_gpu_reqs,
+    translate_gpu_container_resources,
+    translate_gpu_resources,
+    translate_pod_gpu_resources,
+)
+
+
+def test_set_gpu_reqs_takes_max():
+    c = ContainerInfo(
+        requests={RESOURCE_GPU: 2},
+        kube_requests={RESOURCE_GPU: 3},
+    )
+    assert set_gpu_reqs(c) == 3
+    assert c.dev_requests[RESOURCE_GPU] == 3
+
+
+def test_flat_expansion_two_level():
+    """Flat amd.com/gpu: 2 expands to wildcard-wrapped per-card names."""
+    c = ContainerInfo(kube_requests={RESOURCE_GPU: 2})
+    translate_gpu_container_resources(TWO_LEVEL_TEMPLATE, c)
+    assert RESOURCE_GPU not in c.dev_requests
+    assert c.dev_requests == {
+        "resource/group/gpugrp1/*/gpugrp0/*/gpu/0/cards": 1,
+        "resource/group/gpugrp1/*/gpugrp0/*/gpu/1/cards": 1,
+    }
+
+
+def test_flat_expansion_flat_node():
+    """Against a flat (non-grouped) template, names stay ungrouped."""
+    c = ContainerInfo(kube_requests={RESOURCE_GPU: 1})
+    translate_gpu_container_resources({}, c)
+    assert c.dev_requests == {"resource/group/gpu/0/cards": 1}
+
+
+def test_node_allocatable_forcing():
+    """A flat node's allocatable is forced to the 2-level scheme
+    (cf. gpu_scheduler.go:21-28)."""
+    alloc = {RESOURCE_GPU: 2}
+    translate_gpu_resources(2, TWO_LEVEL_TEMPLATE, alloc)
+    assert set(alloc) == {
+        "resource/group/gpugrp1/*/gpugrp0/*/gpu/0/cards",
+        "resource/group/gpugrp1/*/gpugrp0/*/gpu/1/cards",
+    }
+
+
+def _cached_node(cache: NodeTreeCache):
+    res = {}
+    for g in range(8):
+        res[f"resource/group/gpugrp1/0/gpugrp0/0/gpu/GPU{g}/cards"] = 1
+    cache.add_node_resources("node0", res)
+
+
+def test_topology_knob_dispatch():
+    cache = NodeTreeCache()
+    _cached_node(cache)
+    ni = NodeInfo(name="node0")
+
+    # knob unset -> topology synthesis
+    pod = PodInfo(
+        name="p",
+        running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: 2})},
+    )
+    translate_pod_gpu_resources(ni, pod, cache)
+    names = sorted(pod.running_containers["c"].dev_requests)
+    assert names == [
+        "resource/group/gpugrp1/0/gpugrp0/0/gpu/0/cards",
+        "resource/group/gpugrp1/0/gpugrp0/0/gpu/1/cards",
+    ]
+
+    # knob = 0 -> flat path
+    pod0 = PodInfo(
+        name="p0",
+        requests={GPU_TOPOLOGY_GENERATION: 0},
+        running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: 1})},
+    )
+    translate_pod_gpu_resources(ni, pod0, cache)
+    assert list(pod0.running_containers["c"].dev_requests) == [
+        "resource/group/gpu/0/cards"
+    ]
+
+    # invalid knob -> error (gpu.go:102-126)
+    podx = PodInfo(
+        name="px",
+        requests={GPU_TOPOLOGY_GENERATION: 7},
+        running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: 1})},
+    )
+    with pytest.raises(SchedulingError):
+        translate_pod_gpu_resources(ni, podx, cache)
+
+
+def test_no_tree_big_pod_fails():
+    cache = NodeTreeCache()
+    _cached_node(cache)
+    ni = NodeInfo(name="node0")
+    pod = PodInfo(
+        name="big",
+        running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: 9})},
+    )
+    with pytest.raises(SchedulingError):
+        translate_pod_gpu_resources(ni, pod, cache)
+
+
+def test_multi_container_disjoint_slots():
+    cache = NodeTreeCache()
+    _cached_node(cache)
+    ni = NodeInfo(name="node0")
+    pod = PodInfo(
+        name="p",
+        running_containers={
+            "a": ContainerInfo(kube_requests={RESOURCE_GPU: 2}),
+            "b": ContainerInfo(kube_requests={RESOURCE_GPU: 3}),
+        },
+        init_containers={"i": ContainerInfo(kube_requests={RESOURCE_GPU: 2})},
+    )
+    translate_pod_gpu_resources(ni, pod, cache)
+    a = set(pod.running_containers["a"].dev_requests)
+    b = set(pod.running_containers["b"].dev_requests)
+    assert not a & b
+    assert len(a) == 2 and len(b) == 3
+    # init containers restart from slot 0 (max-over-init accounting)
+    i = sorted(pod.init_containers["i"].dev_requests)
+    assert i[0].endswith("/gpu/0/cards")
