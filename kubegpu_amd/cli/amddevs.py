@@ -1,0 +1,79 @@
+"""amddevs — operator CLI (analog of the reference's nvidiadevs,
+/root/reference/nvidiagpuplugin/cmd/main.go:13-45).
+
+Modes:
+  amddevs                 raw discovery dump (GpusInfo JSON)
+  amddevs --plugin        full device-plugin path: New/Start/UpdateNodeInfo,
+                          print the advertised NodeInfo
+  amddevs --schedule K    schedule a synthetic K-GPU pod against the local
+                          node and print the chosen GPU set + allocation
+  amddevs --probe K       same, then run the in-pod RCCL probe over the set
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import sys
+
+from ..api.types import ContainerInfo, NodeInfo, PodInfo
+from ..core import Cluster
+from ..deviceplugin import create_device_plugin
+from ..discovery import default_backend
+from ..plugintypes import RESOURCE_GPU
+
+
+def main(argv=None) -> int:
+    p = argparse.ArgumentParser(prog="amddevs")
+    p.add_argument("--plugin", action="store_true", help="run the device-plugin path")
+    p.add_argument("--schedule", type=int, metavar="K", help="schedule a K-GPU pod")
+    p.add_argument("--probe", type=int, metavar="K", help="schedule + RCCL probe")
+    p.add_argument("--bytes", type=int, default=256 << 20)
+    args = p.parse_args(argv)
+
+    backend = default_backend()
+    if not (args.plugin or args.schedule or args.probe):
+        print(backend.get_gpu_info().decode())
+        return 0
+
+    mgr = create_device_plugin(backend)
+    mgr.start()
+    ni = NodeInfo(name="local")
+    mgr.update_node_info(ni)
+    if args.plugin:
+        print(json.dumps({
+            "capacity": ni.capacity,
+            "allocatable": ni.allocatable,
+            "kube_cap": ni.kube_cap,
+            "kube_alloc": ni.kube_alloc,
+        }, indent=1))
+        return 0
+
+    k = args.probe or args.schedule
+    cluster = Cluster()
+    cluster.add_node(ni, mgr._last_info, mgr)
+    pod = PodInfo(
+        name=f"cli-{k}gpu",
+        running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: k})},
+    )
+    res = cluster.schedule(pod)
+    mounts, devices, envs = cluster.container_allocate(pod, "c")
+    print(json.dumps({
+        "node": res.node_name,
+        "gpus": res.uuids,
+        "devices": devices,
+        "envs": envs,
+        "schedule_latency_ms": res.latency_s * 1e3,
+    }, indent=1))
+
+    if args.probe:
+        from ..probe import run_rccl_probe
+
+        idxs = sorted(mgr.gpus[u].index for u in res.uuids)
+        out = run_rccl_probe(devices=idxs, nbytes=args.bytes)
+        print(json.dumps(out, indent=1))
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

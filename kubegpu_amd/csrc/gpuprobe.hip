@@ -1,0 +1,158 @@
+// gpuprobe — hand-written CDNA4 (gfx950) bandwidth kernels.
+//
+// Used by the probe subsystem for the k=1 degenerate point of the
+// BASELINE.md curve (single-GPU "all-reduce" has no interconnect: the
+// meaningful sanity number is HBM3E streaming bandwidth, spec 8 TB/s,
+// ≈6.3 TB/s achievable per MI355X_MICROARCH.md) and by GPU numerics
+// tests.
+//
+// Kernel notes (per /opt/skills/guides/cdna_hip_programming.md):
+//  * 256-thread blocks = 4 wave64s; 16 B/lane vectorized access
+//    (uint4) -> 4 KiB per block per instruction.
+//  * grid-stride with >> 256 workgroups so all 8 XCDs / 256 CUs fill.
+//  * copy is the float4-copy pattern that measures ~79% of HBM peak.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+#define WAVE 64
+#define BLOCK 256
+
+__global__ void copy_kernel_v4(const uint4* __restrict__ src,
+                               uint4* __restrict__ dst, size_t n4) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n4; i += stride) dst[i] = src[i];
+}
+
+__global__ void copy_kernel_b(const unsigned char* __restrict__ src,
+                              unsigned char* __restrict__ dst, size_t n) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) dst[i] = src[i];
+}
+
+// Read-only streaming: block-sum of uint4 lanes into one u64 per block
+// (measures pure HBM read bandwidth; the result write is negligible).
+__global__ void read_sum_kernel(const uint4* __restrict__ src, size_t n4,
+                                unsigned long long* __restrict__ out) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  unsigned long long acc = 0;
+  for (; i < n4; i += stride) {
+    uint4 v = src[i];
+    acc += (unsigned long long)v.x + v.y + v.z + v.w;
+  }
+  __shared__ unsigned long long sh[BLOCK / WAVE];
+  // wave-level reduce via shuffles (64-wide waves on CDNA)
+  for (int off = WAVE / 2; off > 0; off >>= 1)
+    acc += __shfl_down(acc, off, WAVE);
+  int lane = threadIdx.x % WAVE;
+  int wid = threadIdx.x / WAVE;
+  if (lane == 0) sh[wid] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    unsigned long long total = 0;
+    for (int w = 0; w < BLOCK / WAVE; ++w) total += sh[w];
+    atomicAdd(out, total);
+  }
+}
+
+static void check_pair(const at::Tensor& dst, const at::Tensor& src) {
+  TORCH_CHECK(src.is_cuda() && dst.is_cuda(), "gpuprobe: tensors must be on GPU");
+  TORCH_CHECK(src.is_contiguous() && dst.is_contiguous(),
+              "gpuprobe: tensors must be contiguous");
+  TORCH_CHECK(src.nbytes() == dst.nbytes(), "gpuprobe: size mismatch");
+}
+
+void copy(at::Tensor dst, at::Tensor src) {
+  check_pair(dst, src);
+  size_t nbytes = src.nbytes();
+  auto stream = at::hip::getCurrentHIPStream();
+  if (nbytes % 16 == 0) {
+    size_t n4 = nbytes / 16;
+    int blocks = (int)std::min<size_t>((n4 + BLOCK - 1) / BLOCK, 4096);
+    hipLaunchKernelGGL(copy_kernel_v4, dim3(blocks), dim3(BLOCK), 0, stream,
+                       (const uint4*)src.data_ptr(), (uint4*)dst.data_ptr(), n4);
+  } else {
+    int blocks = (int)std::min<size_t>((nbytes + BLOCK - 1) / BLOCK, 4096);
+    hipLaunchKernelGGL(copy_kernel_b, dim3(blocks), dim3(BLOCK), 0, stream,
+                       (const unsigned char*)src.data_ptr(),
+                       (unsigned char*)dst.data_ptr(), nbytes);
+  }
+  C10_HIP_KERNEL_LAUNCH_CHECK();
+}
+
+// Timed device-to-device streaming copy; returns achieved GB/s
+// (bytes read + bytes written over wall time, hipEvent-timed).
+double copy_bw_gbps(int64_t nbytes, int64_t iters) {
+  TORCH_CHECK(nbytes > 0 && nbytes % 16 == 0, "nbytes must be positive, 16-aligned");
+  auto opts = at::TensorOptions().dtype(at::kByte).device(at::kCUDA);
+  at::Tensor src = at::empty({nbytes}, opts);
+  at::Tensor dst = at::empty({nbytes}, opts);
+  src.fill_(1);
+  auto stream = at::hip::getCurrentHIPStream();
+  size_t n4 = (size_t)nbytes / 16;
+  int blocks = (int)std::min<size_t>((n4 + BLOCK - 1) / BLOCK, 4096);
+  // warmup
+  for (int w = 0; w < 3; ++w)
+    hipLaunchKernelGGL(copy_kernel_v4, dim3(blocks), dim3(BLOCK), 0, stream,
+                       (const uint4*)src.data_ptr(), (uint4*)dst.data_ptr(), n4);
+  hipEvent_t t0, t1;
+  (void)hipEventCreate(&t0);
+  (void)hipEventCreate(&t1);
+  (void)hipEventRecord(t0, stream);
+  for (int64_t i = 0; i < iters; ++i)
+    hipLaunchKernelGGL(copy_kernel_v4, dim3(blocks), dim3(BLOCK), 0, stream,
+                       (const uint4*)src.data_ptr(), (uint4*)dst.data_ptr(), n4);
+  (void)hipEventRecord(t1, stream);
+  (void)hipEventSynchronize(t1);
+  float ms = 0.f;
+  (void)hipEventElapsedTime(&ms, t0, t1);
+  (void)hipEventDestroy(t0);
+  (void)hipEventDestroy(t1);
+  C10_HIP_KERNEL_LAUNCH_CHECK();
+  double sec = ms / 1e3;
+  return (double)nbytes * 2.0 * iters / sec / 1e9;
+}
+
+double read_bw_gbps(int64_t nbytes, int64_t iters) {
+  TORCH_CHECK(nbytes > 0 && nbytes % 16 == 0, "nbytes must be positive, 16-aligned");
+  auto opts = at::TensorOptions().dtype(at::kByte).device(at::kCUDA);
+  at::Tensor src = at::empty({nbytes}, opts);
+  src.fill_(1);
+  at::Tensor out = at::zeros({1}, at::TensorOptions().dtype(at::kLong).device(at::kCUDA));
+  auto stream = at::hip::getCurrentHIPStream();
+  size_t n4 = (size_t)nbytes / 16;
+  int blocks = (int)std::min<size_t>((n4 + BLOCK - 1) / BLOCK, 4096);
+  for (int w = 0; w < 3; ++w)
+    hipLaunchKernelGGL(read_sum_kernel, dim3(blocks), dim3(BLOCK), 0, stream,
+                       (const uint4*)src.data_ptr(), n4,
+                       (unsigned long long*)out.data_ptr());
+  hipEvent_t t0, t1;
+  (void)hipEventCreate(&t0);
+  (void)hipEventCreate(&t1);
+  (void)hipEventRecord(t0, stream);
+  for (int64_t i = 0; i < iters; ++i)
+    hipLaunchKernelGGL(read_sum_kernel, dim3(blocks), dim3(BLOCK), 0, stream,
+                       (const uint4*)src.data_ptr(), n4,
+                       (unsigned long long*)out.data_ptr());
+  (void)hipEventRecord(t1, stream);
+  (void)hipEventSynchronize(t1);
+  float ms = 0.f;
+  (void)hipEventElapsedTime(&ms, t0, t1);
+  (void)hipEventDestroy(t0);
+  (void)hipEventDestroy(t1);
+  C10_HIP_KERNEL_LAUNCH_CHECK();
+  double sec = ms / 1e3;
+  return (double)nbytes * iters / sec / 1e9;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("copy", &copy, "streaming uint4 copy kernel (dst, src)");
+  m.def("copy_bw_gbps", &copy_bw_gbps, "timed d2d copy bandwidth",
+        py::arg("nbytes"), py::arg("iters") = 20);
+  m.def("read_bw_gbps", &read_bw_gbps, "timed read bandwidth",
+        py::arg("nbytes"), py::arg("iters") = 20);
+}

@@ -1,0 +1,4 @@
+"""Bandwidth probes: RCCL-over-xGMI all-reduce + CDNA4 HBM kernels."""
+
+from .bandwidth import copy, d2d_copy_bw_gbps, load_ext, read_bw_gbps  # noqa: F401
+from .rccl_probe import run_rccl_probe, torch_allreduce_busbw  # noqa: F401

@@ -1,0 +1,78 @@
+"""Multi-process distributed-path tests (gloo backend, CPU).
+
+The 8-GPU scaling bench is driver-run; these tests keep the distributed
+code path (torch.distributed init, bucketed all-reduce probe, MAX-over-
+ranks timing) correct by construction on CPU with world_size=2.
+"""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _run_dist(script: str, nproc: int = 2, timeout: int = 240):
+    env = dict(os.environ)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    return subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", f"--nproc-per-node={nproc}",
+            "--master-addr", "127.0.0.1", "--master-port", "29591",
+            script, *(["--gpus", str(nproc), "--steps", "3", "--warmup", "1",
+                       "--pods", "50"] if script.endswith("bench.py") else []),
+        ],
+        capture_output=True,
+        timeout=timeout,
+        env=env,
+        cwd=REPO,
+        text=True,
+    )
+
+
+@pytest.mark.timeout(300)
+def test_bench_world2_gloo_contract():
+    """bench.py over 2 gloo ranks prints one valid JSON contract line."""
+    res = _run_dist(os.path.join(REPO, "bench.py"))
+    assert res.returncode == 0, res.stderr[-2000:]
+    json_lines = [
+        l for l in res.stdout.splitlines() if l.startswith("{") and '"metric"' in l
+    ]
+    assert len(json_lines) == 1  # rank 0 only
+    rec = json.loads(json_lines[0])
+    assert rec["n_gpus"] == 2
+    assert rec["steps"] == 3
+    assert rec["higher_is_better"] is True
+    assert rec["scaling"] == "weak"
+    assert rec["dtype"] == "bf16"
+    assert rec["data"] == "synthetic"
+    assert rec["value"] > 0
+    assert "schedule_p50_ms" in rec
+    assert rec["config"]["mode"] == "rccl_allreduce"
+
+
+HELPER = """
+import os, sys
+sys.path.insert(0, {repo!r})
+import torch.distributed as dist
+from kubegpu_amd.probe.rccl_probe import torch_allreduce_busbw
+dist.init_process_group(backend="gloo")
+out = torch_allreduce_busbw(nbytes=1 << 20, iters=4, warmup=1)
+if dist.get_rank() == 0:
+    assert out["world"] == 2 and out["busbw_gbps"] > 0
+    print("PROBE_OK", out["busbw_gbps"])
+dist.destroy_process_group()
+"""
+
+
+@pytest.mark.timeout(300)
+def test_torch_allreduce_probe_world2(tmp_path):
+    script = tmp_path / "probe_helper.py"
+    script.write_text(HELPER.format(repo=REPO))
+    res = _run_dist(str(script))
+    assert res.returncode == 0, res.stderr[-2000:]
+    assert "PROBE_OK" in res.stdout
