@@ -25,7 +25,6 @@ from ..deviceplugin.manager import AMDGPUManager
 from ..discovery import GpusInfo
 from ..scheduler.scheduler import AMDGPUScheduler
 from ..scheduler.translate import SchedulingError
-from ..scheduler.xgmi import best_ring, xgmi_edges
 from .group_scheduler import GroupScheduler
 
 
@@ -83,9 +82,9 @@ class Cluster:
         candidates: List[Tuple[Tuple[float, int], str, PodInfo, List[str]]] = []
         for name in utils.sorted_string_keys(self.node_infos):
             ni = self.node_infos[name]
-            fits, _, _ = self.scheduler.pod_fits_device(ni, pod, fill_allocate_from=False)
-            if not fits:
-                continue
+            # fit == "a translation + binding exists": the bind attempt
+            # below subsumes the pod_fits_device predicate (which stays
+            # available for API parity / external callers).
             trial = pod.copy()
             try:
                 self.scheduler.pod_allocate(ni, trial)
@@ -94,12 +93,12 @@ class Cluster:
                 continue
             state = self.core.nodes[name]
             idxs = [state.gpus[u].index for u in uuids]
-            ring_bw, _ = best_ring(idxs, state.bw) if idxs else (0.0, [])
+            ring_bw = state.scorer.ring_bw(idxs) if idxs else 0.0
             remaining = [
                 state.gpus[u].index for u in state.free_uuids() if u not in set(uuids)
             ]
-            frag = xgmi_edges(remaining, state.bw)
-            score = (min(ring_bw, 1e9), frag)
+            frag = state.scorer.edges(remaining)
+            score = (ring_bw, frag)
             candidates.append((score, name, trial, uuids))
         if not candidates:
             raise SchedulingError(f"no node fits pod {pod.name}")

@@ -36,7 +36,7 @@ from ..discovery import (
 )
 from ..scheduler.translate import SchedulingError
 from ..scheduler.treecache import LabeledLayout, parse_node_resources
-from ..scheduler.xgmi import BwMatrix, best_ring, choose_best_subset_fast
+from ..scheduler.xgmi import BwMatrix, TopologyScorer, best_ring
 
 
 def _ring_of(indices, bw):
@@ -62,6 +62,7 @@ class NodeState:
     used: Set[str] = field(default_factory=set)  # uuids
     bw: BwMatrix = field(default_factory=dict)  # index -> index -> GB/s
     index_to_uuid: Dict[int, str] = field(default_factory=dict)
+    scorer: Optional[TopologyScorer] = None
 
     def free_uuids(self) -> List[str]:
         return [u for u in sorted(self.gpus) if u not in self.used]
@@ -127,6 +128,7 @@ class GroupScheduler:
             state.bw = gpus_info.bandwidth_matrix()
         else:
             state.bw = _synthetic_bw(list(state.gpus.values()))
+        state.scorer = TopologyScorer(list(state.index_to_uuid.keys()), state.bw)
         with self._lock:
             self.nodes[node_info.name] = state
         return state
@@ -192,7 +194,7 @@ class GroupScheduler:
                     members = self._group_members(state, ah, ag)
                     cand = [state.gpus[u].index for u in members if u in free]
                     k = len(demands[(hi, gi)])
-                    picked = choose_best_subset_fast(cand, k, state.bw)
+                    picked = state.scorer.choose(cand, k)
                     picked_uuids = [state.index_to_uuid[i] for i in picked]
                     for (cont, req), uuid in zip(demands[(hi, gi)], picked_uuids):
                         bindings.append((cont, req, state.gpus[uuid].concrete_name))
@@ -204,7 +206,7 @@ class GroupScheduler:
             if wkey in demands:
                 k = len(demands[wkey])
                 cand = [state.gpus[u].index for u in sorted(free)]
-                picked = choose_best_subset_fast(cand, k, state.bw)
+                picked = state.scorer.choose(cand, k)
                 if len(picked) < k:
                     raise SchedulingError(
                         f"node {node_name}: {len(cand)} free GPUs, pod "
@@ -286,9 +288,8 @@ class GroupScheduler:
                 used_ag.add(ag)
                 out[gi] = ag
                 idxs = [state.gpus[u].index for u in free_per[(ah, ag)]]
-                picked = choose_best_subset_fast(idxs, count, state.bw)
-                bw_val, _ = _ring_of(picked, state.bw)
-                quality += bw_val
+                picked = state.scorer.choose(idxs, count)
+                quality += state.scorer.ring_bw(picked)
                 leftover += n_free - count
             return out, (quality, -leftover)
 

@@ -71,9 +71,16 @@ std::string json_escape(const std::string& s) {
 
 std::string gfx_name(uint64_t v) {
   if (v == 0 || v == UINT64_MAX) return "";
-  uint64_t major = v / 10000, minor = (v / 100) % 100, step = v % 100;
   char buf[32];
-  snprintf(buf, sizeof(buf), "gfx%" PRIu64 "%" PRIx64 "%" PRIx64, major, minor, step);
+  if (v < 0x10000) {
+    // amdsmi reports the target as a hex-coded literal: 0x950 -> gfx950
+    // (measured on MI355 hardware; 0x75a3 device reports 0x950 here).
+    snprintf(buf, sizeof(buf), "gfx%" PRIx64, v);
+  } else {
+    // decimal KFD encoding: major*10000 + minor*100 + step (e.g. 90500)
+    uint64_t major = v / 10000, minor = (v / 100) % 100, step = v % 100;
+    snprintf(buf, sizeof(buf), "gfx%" PRIu64 "%" PRIx64 "%" PRIx64, major, minor, step);
+  }
   return buf;
 }
 
@@ -260,10 +267,39 @@ void print_human(const std::vector<Gpu>& gpus, const std::string& driver) {
   }
 }
 
+// xGMI per-link traffic counters (read/write KB since boot), one JSON
+// line per call.  The probe wrapper diffs two snapshots around an
+// all-reduce to report per-link utilization (BASELINE.md "xGMI link
+// utilization during probe"; counters via amdsmi_get_link_metrics).
+int print_link_metrics(const std::vector<Gpu>& gpus) {
+  printf("{\"gpus\": [");
+  for (size_t i = 0; i < gpus.size(); ++i) {
+    const Gpu& g = gpus[i];
+    amdsmi_link_metrics_t lm{};
+    printf("%s{\"index\": %d, \"links\": [", i ? "," : "", g.index);
+    if (amdsmi_get_link_metrics(g.handle, &lm) == AMDSMI_STATUS_SUCCESS) {
+      int emitted = 0;
+      for (uint32_t l = 0; l < lm.num_links && l < AMDSMI_MAX_NUM_XGMI_PHYSICAL_LINK; ++l) {
+        const auto& lk = lm.links[l];
+        printf("%s{\"link\": %u, \"type\": \"%s\", \"bit_rate_gbps\": %u, "
+               "\"max_bandwidth_gbps\": %u, \"read_kb\": %" PRIu64
+               ", \"write_kb\": %" PRIu64 "}",
+               emitted ? "," : "", l, link_type_name(lk.link_type), lk.bit_rate,
+               lk.max_bandwidth, lk.read, lk.write);
+        ++emitted;
+      }
+    }
+    printf("]}");
+  }
+  printf("]}\n");
+  return 0;
+}
+
 }  // namespace
 
 int main(int argc, char** argv) {
-  bool as_json = argc > 1 && std::string(argv[1]) == "json";
+  std::string mode = argc > 1 ? argv[1] : "";
+  bool as_json = mode == "json";
   if (amdsmi_init(AMDSMI_INIT_AMD_GPUS) != AMDSMI_STATUS_SUCCESS) {
     fprintf(stderr, "amdsmiinfo: amdsmi_init failed\n");
     return 1;
@@ -276,10 +312,13 @@ int main(int argc, char** argv) {
     amdsmi_shut_down();
     return 1;
   }
+  int rc = 0;
   if (as_json)
     print_json(gpus, driver);
+  else if (mode == "linkmetrics")
+    rc = print_link_metrics(gpus);
   else
     print_human(gpus, driver);
   amdsmi_shut_down();
-  return 0;
+  return rc;
 }

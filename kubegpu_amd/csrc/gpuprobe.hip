@@ -26,6 +26,21 @@ __global__ void copy_kernel_v4(const uint4* __restrict__ src,
   for (; i < n4; i += stride) dst[i] = src[i];
 }
 
+// Non-temporal variant: bypasses L2/LLC allocation on both sides, which
+// matters for pure streaming (the 256 MiB Infinity Cache otherwise
+// absorbs part of the stream and skews small-buffer numbers).
+typedef unsigned int uint4v __attribute__((ext_vector_type(4)));
+
+__global__ void copy_kernel_v4_nt(const uint4v* __restrict__ src,
+                                  uint4v* __restrict__ dst, size_t n4) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n4; i += stride) {
+    uint4v v = __builtin_nontemporal_load(&src[i]);
+    __builtin_nontemporal_store(v, &dst[i]);
+  }
+}
+
 __global__ void copy_kernel_b(const unsigned char* __restrict__ src,
                               unsigned char* __restrict__ dst, size_t n) {
   size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -86,7 +101,9 @@ void copy(at::Tensor dst, at::Tensor src) {
 
 // Timed device-to-device streaming copy; returns achieved GB/s
 // (bytes read + bytes written over wall time, hipEvent-timed).
-double copy_bw_gbps(int64_t nbytes, int64_t iters) {
+// blocks=0 picks the default; nontemporal selects the NT variant.
+double copy_bw_gbps(int64_t nbytes, int64_t iters, int64_t blocks_arg,
+                    bool nontemporal) {
   TORCH_CHECK(nbytes > 0 && nbytes % 16 == 0, "nbytes must be positive, 16-aligned");
   auto opts = at::TensorOptions().dtype(at::kByte).device(at::kCUDA);
   at::Tensor src = at::empty({nbytes}, opts);
@@ -94,18 +111,23 @@ double copy_bw_gbps(int64_t nbytes, int64_t iters) {
   src.fill_(1);
   auto stream = at::hip::getCurrentHIPStream();
   size_t n4 = (size_t)nbytes / 16;
-  int blocks = (int)std::min<size_t>((n4 + BLOCK - 1) / BLOCK, 4096);
-  // warmup
-  for (int w = 0; w < 3; ++w)
-    hipLaunchKernelGGL(copy_kernel_v4, dim3(blocks), dim3(BLOCK), 0, stream,
-                       (const uint4*)src.data_ptr(), (uint4*)dst.data_ptr(), n4);
+  int blocks = blocks_arg > 0
+                   ? (int)blocks_arg
+                   : (int)std::min<size_t>((n4 + BLOCK - 1) / BLOCK, 4096);
+  auto launch = [&]() {
+    if (nontemporal)
+      hipLaunchKernelGGL(copy_kernel_v4_nt, dim3(blocks), dim3(BLOCK), 0, stream,
+                         (const uint4v*)src.data_ptr(), (uint4v*)dst.data_ptr(), n4);
+    else
+      hipLaunchKernelGGL(copy_kernel_v4, dim3(blocks), dim3(BLOCK), 0, stream,
+                         (const uint4*)src.data_ptr(), (uint4*)dst.data_ptr(), n4);
+  };
+  for (int w = 0; w < 3; ++w) launch();
   hipEvent_t t0, t1;
   (void)hipEventCreate(&t0);
   (void)hipEventCreate(&t1);
   (void)hipEventRecord(t0, stream);
-  for (int64_t i = 0; i < iters; ++i)
-    hipLaunchKernelGGL(copy_kernel_v4, dim3(blocks), dim3(BLOCK), 0, stream,
-                       (const uint4*)src.data_ptr(), (uint4*)dst.data_ptr(), n4);
+  for (int64_t i = 0; i < iters; ++i) launch();
   (void)hipEventRecord(t1, stream);
   (void)hipEventSynchronize(t1);
   float ms = 0.f;
@@ -152,7 +174,8 @@ double read_bw_gbps(int64_t nbytes, int64_t iters) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("copy", &copy, "streaming uint4 copy kernel (dst, src)");
   m.def("copy_bw_gbps", &copy_bw_gbps, "timed d2d copy bandwidth",
-        py::arg("nbytes"), py::arg("iters") = 20);
+        py::arg("nbytes"), py::arg("iters") = 20, py::arg("blocks") = 0,
+        py::arg("nontemporal") = false);
   m.def("read_bw_gbps", &read_bw_gbps, "timed read bandwidth",
         py::arg("nbytes"), py::arg("iters") = 20);
 }

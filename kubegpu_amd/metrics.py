@@ -12,7 +12,13 @@ import threading
 from typing import Dict, List, Optional
 
 try:
-    from prometheus_client import Counter, Gauge, Histogram, start_http_server
+    from prometheus_client import (
+        CollectorRegistry,
+        Counter,
+        Gauge,
+        Histogram,
+        start_http_server,
+    )
 
     _HAVE_PROM = True
 except ImportError:  # pragma: no cover
@@ -20,21 +26,32 @@ except ImportError:  # pragma: no cover
 
 
 class Metrics:
-    def __init__(self) -> None:
+    def __init__(self, registry=None) -> None:
         self._lock = threading.Lock()
         self.schedule_latencies: List[float] = []
         self.allocations = 0
         self.failures = 0
         if _HAVE_PROM:
+            # per-instance registry: multiple Metrics objects (tests,
+            # embedded schedulers) must not collide in the global one
+            self.registry = registry if registry is not None else CollectorRegistry()
             self._h = Histogram(
                 "kubegpu_amd_schedule_latency_seconds",
                 "pod schedule latency",
                 buckets=(1e-5, 1e-4, 5e-4, 1e-3, 5e-3, 1e-2, 0.1, 1.0),
+                registry=self.registry,
             )
-            self._alloc = Counter("kubegpu_amd_allocations_total", "pod GPU allocations")
-            self._fail = Counter("kubegpu_amd_schedule_failures_total", "schedule failures")
+            self._alloc = Counter(
+                "kubegpu_amd_allocations_total", "pod GPU allocations",
+                registry=self.registry,
+            )
+            self._fail = Counter(
+                "kubegpu_amd_schedule_failures_total", "schedule failures",
+                registry=self.registry,
+            )
             self._xgmi = Gauge(
-                "kubegpu_amd_xgmi_link_gbps", "last probed xGMI ring bandwidth GB/s"
+                "kubegpu_amd_xgmi_link_gbps", "last probed xGMI ring bandwidth GB/s",
+                registry=self.registry,
             )
 
     def observe_schedule(self, seconds: float) -> None:
@@ -69,7 +86,7 @@ class Metrics:
 
     def serve(self, port: int = 9400) -> bool:
         if _HAVE_PROM:
-            start_http_server(port)
+            start_http_server(port, registry=self.registry)
             return True
         return False
 

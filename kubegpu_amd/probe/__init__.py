@@ -2,3 +2,8 @@
 
 from .bandwidth import copy, d2d_copy_bw_gbps, load_ext, read_bw_gbps  # noqa: F401
 from .rccl_probe import run_rccl_probe, torch_allreduce_busbw  # noqa: F401
+from .xgmi_counters import (  # noqa: F401
+    diff_link_metrics,
+    probe_with_link_utilization,
+    read_link_metrics,
+)

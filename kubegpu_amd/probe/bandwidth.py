@@ -51,9 +51,14 @@ def copy(dst, src) -> None:
     load_ext().copy(dst, src)
 
 
-def d2d_copy_bw_gbps(nbytes: int = 1 << 30, iters: int = 20) -> float:
+def d2d_copy_bw_gbps(
+    nbytes: int = 1 << 30,
+    iters: int = 20,
+    blocks: int = 0,
+    nontemporal: bool = False,
+) -> float:
     """Timed device-to-device streaming-copy bandwidth (GB/s, R+W)."""
-    return load_ext().copy_bw_gbps(nbytes, iters)
+    return load_ext().copy_bw_gbps(nbytes, iters, blocks, nontemporal)
 
 
 def read_bw_gbps(nbytes: int = 1 << 30, iters: int = 20) -> float:

@@ -153,6 +153,85 @@ def choose_best_subset(
     return best_sub
 
 
+class TopologyScorer:
+    """Per-node memoized subset scorer.
+
+    Holds the node's symmetrized bandwidth matrix flattened once (the
+    per-call rebuild dominated the schedule path at ~0.75 ms/pod) and
+    memoizes subset choices and ring bottlenecks by (free-set, k) — the
+    topology is static per node, so steady-state pod streams hit the
+    memo almost always.  Uses the native _schedcore twin when built.
+    """
+
+    def __init__(self, indices: Sequence[int], bw: BwMatrix):
+        self.idx = sorted(set(indices))
+        self.n = len(self.idx)
+        self.pos = {g: i for i, g in enumerate(self.idx)}
+        self.flat = [0.0] * (self.n * self.n)
+        for a in range(self.n):
+            for b in range(self.n):
+                if a != b:
+                    self.flat[a * self.n + b] = _sym_bw(bw, self.idx[a], self.idx[b])
+        self._native = None
+        if not os.environ.get("KUBEGPU_PURE_PY"):
+            try:
+                from .. import _schedcore
+
+                self._native = _schedcore
+            except ImportError:
+                pass
+        self._bw = bw
+        self._choose_memo: Dict[Tuple, List[int]] = {}
+        self._ring_memo: Dict[Tuple, float] = {}
+        self._edges_memo: Dict[Tuple, int] = {}
+
+    def choose(self, free: Sequence[int], k: int) -> List[int]:
+        key = (tuple(sorted(free)), k)
+        hit = self._choose_memo.get(key)
+        if hit is not None:
+            return list(hit)
+        free_sorted = sorted(free)
+        if k <= 0 or k > len(free_sorted):
+            out: List[int] = []
+        elif self._native is not None:
+            if len(free_sorted) == self.n:
+                picked = self._native.choose_best_subset(self.n, k, self.flat)
+                out = [self.idx[p] for p in picked]
+            else:
+                # sub-matrix for the current free set
+                m = len(free_sorted)
+                sub = [0.0] * (m * m)
+                for a in range(m):
+                    pa = self.pos[free_sorted[a]]
+                    for b in range(m):
+                        if a != b:
+                            sub[a * m + b] = self.flat[pa * self.n + self.pos[free_sorted[b]]]
+                picked = self._native.choose_best_subset(m, k, sub)
+                out = [free_sorted[p] for p in picked]
+        else:
+            out = choose_best_subset(free_sorted, k, self._bw)
+        self._choose_memo[key] = out
+        return list(out)
+
+    def ring_bw(self, subset: Sequence[int]) -> float:
+        key = tuple(sorted(subset))
+        hit = self._ring_memo.get(key)
+        if hit is not None:
+            return hit
+        val, _ = best_ring(sorted(subset), self._bw)
+        val = min(val, 1e9)
+        self._ring_memo[key] = val
+        return val
+
+    def edges(self, gpus: Sequence[int]) -> int:
+        key = tuple(sorted(gpus))
+        hit = self._edges_memo.get(key)
+        if hit is None:
+            hit = xgmi_edges(gpus, self._bw)
+            self._edges_memo[key] = hit
+        return hit
+
+
 def _native_available() -> bool:
     if os.environ.get("KUBEGPU_PURE_PY"):
         return False

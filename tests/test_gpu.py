@@ -37,7 +37,7 @@ def test_amdsmiinfo_binary_json():
     assert len(info.devices) >= 1
     g = info.devices[0]
     assert g.uuid
-    assert g.gfx_target.startswith("gfx95")
+    assert g.gfx_target == "gfx950"
     # MI355X: 288 GB HBM3E
     assert g.memory.vram_total_bytes > 200 * 1024**3
     assert g.render_path.startswith("/dev/dri/renderD")

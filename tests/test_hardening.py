@@ -1,0 +1,169 @@
+"""Hardening tests: concurrency, fault injection, degenerate inputs
+(SURVEY.md §5 / §7 step 6)."""
+
+import json
+import os
+import threading
+
+import pytest
+
+from kubegpu_amd.api.types import ContainerInfo, NodeInfo, PodInfo
+from kubegpu_amd.core import Cluster
+from kubegpu_amd.deviceplugin import create_device_plugin
+from kubegpu_amd.discovery import FakeBackend, SysfsBackend, DiscoveryError, fixtures
+from kubegpu_amd.plugintypes import RESOURCE_GPU
+from kubegpu_amd.probe.xgmi_counters import diff_link_metrics
+from kubegpu_amd.scheduler import NodeTreeCache, SchedulingError
+
+
+def test_concurrent_schedule_release():
+    """Many threads scheduling/releasing against one cluster: no
+    corruption, no double-allocation (the reference's scheduler cache is
+    unsynchronized; ours must not be)."""
+    cluster = Cluster()
+    for n in range(2):
+        mgr = create_device_plugin(FakeBackend(fixtures.fixture_8x_mi355x()))
+        cluster.add_node_from_manager(f"n{n}", mgr)
+    errors = []
+    allocated = []
+    lock = threading.Lock()
+
+    def worker(tid):
+        try:
+            for i in range(30):
+                pod = PodInfo(
+                    name=f"t{tid}-{i}",
+                    running_containers={
+                        "c": ContainerInfo(kube_requests={RESOURCE_GPU: 2})
+                    },
+                )
+                try:
+                    res = cluster.schedule(pod)
+                except SchedulingError:
+                    continue
+                with lock:
+                    allocated.append((res.node_name, tuple(sorted(res.uuids))))
+                cluster.release(pod)
+        except Exception as e:  # pragma: no cover
+            errors.append(e)
+
+    threads = [threading.Thread(target=worker, args=(t,)) for t in range(8)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert not errors
+    # cluster drained back to fully free
+    assert cluster.core.free_count("n0") == 8
+    assert cluster.core.free_count("n1") == 8
+
+
+def test_concurrent_cache_mutation():
+    cache = NodeTreeCache()
+
+    def res_for(shape_id):
+        res = {}
+        for g in range(8):
+            grp = g // (2 if shape_id % 2 else 4)
+            res[f"resource/group/gpugrp1/0/gpugrp0/{grp}/gpu/G{g}/cards"] = 1
+        return res
+
+    def worker(tid):
+        for i in range(200):
+            cache.add_node_resources(f"node{tid}", res_for(i))
+        cache.remove_node(f"node{tid}")
+
+    threads = [threading.Thread(target=worker, args=(t,)) for t in range(8)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert len(cache) == 0
+    assert cache.node_location_map == {}
+
+
+def test_sysfs_backend_missing_root():
+    with pytest.raises(DiscoveryError):
+        SysfsBackend(root="/nonexistent/kfd").get_gpu_info()
+
+
+def test_sysfs_backend_synthetic_tree(tmp_path):
+    """Parse a synthetic KFD topology: 1 CPU node + 2 GPU nodes with an
+    xGMI io_link between them."""
+    def write_node(n, props, links=None, mem=None):
+        d = tmp_path / str(n)
+        d.mkdir()
+        (d / "properties").write_text(
+            "".join(f"{k} {v}\n" for k, v in props.items())
+        )
+        if links:
+            for i, lp in enumerate(links):
+                ld = d / "io_links" / str(i)
+                ld.mkdir(parents=True)
+                (ld / "properties").write_text(
+                    "".join(f"{k} {v}\n" for k, v in lp.items())
+                )
+        if mem:
+            md = d / "mem_banks" / "0"
+            md.mkdir(parents=True)
+            (md / "properties").write_text(
+                "".join(f"{k} {v}\n" for k, v in mem.items())
+            )
+
+    write_node(0, {"simd_count": 0})  # CPU
+    write_node(
+        1,
+        {"simd_count": 1024, "simd_per_cu": 4, "gfx_target_version": 90500,
+         "drm_render_minor": 128, "device_id": 30115, "location_id": 2048,
+         "domain": 0, "unique_id": 111},
+        links=[{"type": 2, "node_to": 2, "weight": 15, "max_bandwidth": 153000}],
+        mem={"heap_type": 1, "size_in_bytes": 309237645312},
+    )
+    write_node(
+        2,
+        {"simd_count": 1024, "simd_per_cu": 4, "gfx_target_version": 90500,
+         "drm_render_minor": 129, "device_id": 30115, "location_id": 2304,
+         "domain": 0, "unique_id": 222},
+        links=[{"type": 2, "node_to": 1, "weight": 15, "max_bandwidth": 153000}],
+        mem={"heap_type": 1, "size_in_bytes": 309237645312},
+    )
+    info = SysfsBackend(root=str(tmp_path)).get_devices()
+    assert len(info.devices) == 2
+    g = info.devices[0]
+    assert g.gfx_target == "gfx950"
+    assert g.render_path == "/dev/dri/renderD128"
+    assert g.memory.vram_total_bytes == 309237645312
+    assert len(g.links) == 1
+    assert g.links[0].type == "XGMI"
+    assert g.links[0].bandwidth_gbps == 153.0
+
+
+def test_link_metrics_diff():
+    before = {0: [{"link": 0, "type": "XGMI", "read_kb": 1000, "write_kb": 2000}]}
+    after = {0: [{"link": 0, "type": "XGMI", "read_kb": 1024000 + 1000,
+                  "write_kb": 2048000 + 2000}]}
+    d = diff_link_metrics(before, after)
+    assert d[0][0]["read_mb"] == 1000.0
+    assert d[0][0]["write_mb"] == 2000.0
+
+
+def test_metrics_percentiles():
+    from kubegpu_amd.metrics import Metrics
+
+    m = Metrics()
+    for v in [0.001, 0.002, 0.003, 0.004]:
+        m.observe_schedule(v)
+    assert m.percentile(0.5) == 0.003
+    m.inc_allocation()
+    m.inc_failure()
+    assert m.allocations == 1 and m.failures == 1
+
+
+def test_zero_gpu_pod_schedules_nowhere_needed():
+    """A pod with no GPU request translates to nothing and needs no bind."""
+    cluster = Cluster()
+    mgr = create_device_plugin(FakeBackend(fixtures.fixture_8x_mi355x()))
+    cluster.add_node_from_manager("n0", mgr)
+    pod = PodInfo(name="nogpu", running_containers={"c": ContainerInfo()})
+    res = cluster.schedule(pod)
+    assert res.uuids == []
