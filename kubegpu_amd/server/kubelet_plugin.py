@@ -1,0 +1,252 @@
+"""Kubelet device-plugin server for MI355X GPUs.
+
+Serves the v1beta1 DevicePlugin gRPC API over a unix socket, backed by
+AMDGPUManager.  This is the stock-Kubernetes path (complementing the
+KubeDevice-style grouped path in kubegpu_amd.core):
+
+* ListAndWatch advertises one kubelet Device per GPU (ID = uuid, NUMA
+  topology attached) under resource ``amd.com/gpu``;
+* GetPreferredAllocation answers kubelet's "which k of these free GPUs?"
+  with the xGMI ring-bandwidth-best subset (scheduler/xgmi.py) — the
+  MI355X topology awareness stock kubelet can actually consume;
+* Allocate returns /dev/kfd + the chosen GPUs' /dev/dri render (+card)
+  nodes as DeviceSpecs plus ROCR_VISIBLE_DEVICES — no vendor runtime
+  hook (north star).
+
+Health: a background loop re-runs discovery; GPUs that vanish flip to
+Unhealthy in the next ListAndWatch frame (mark/sweep semantics of the
+manager, cf. nvidia_gpu_manager.go:132-155).
+"""
+
+from __future__ import annotations
+
+import os
+import queue
+import threading
+import time
+from concurrent import futures
+from typing import Dict, List, Optional
+
+import grpc
+
+from ..api import utils
+from ..deviceplugin.manager import AMDGPUManager
+from ..discovery import DiscoveryError
+from ..plugintypes import RESOURCE_GPU
+from ..scheduler.xgmi import TopologyScorer
+from . import dpapi
+
+
+def _serialize(msg):
+    return msg.SerializeToString()
+
+
+class DevicePluginServicer:
+    def __init__(self, manager: AMDGPUManager, health_interval_s: float = 30.0):
+        self.manager = manager
+        self.health_interval_s = health_interval_s
+        self._stop = threading.Event()
+        self._update = threading.Condition()
+        self._scorer: Optional[TopologyScorer] = None
+
+    # -- device list -------------------------------------------------------
+
+    def _device_list(self) -> List:
+        devices = []
+        try:
+            self.manager.update_gpu_info()
+            healthy = True
+        except DiscoveryError:
+            healthy = False
+        for uuid in sorted(self.manager.gpus):
+            g = self.manager.gpus[uuid]
+            topo = dpapi.TopologyInfo(nodes=[dpapi.NUMANode(ID=g.numa_node)])
+            devices.append(
+                dpapi.Device(
+                    ID=uuid,
+                    health=dpapi.HEALTHY if healthy else dpapi.UNHEALTHY,
+                    topology=topo,
+                )
+            )
+        return devices
+
+    def _refresh_scorer(self) -> Optional[TopologyScorer]:
+        info = self.manager._last_info
+        if info is None or not info.devices:
+            return None
+        bw = info.bandwidth_matrix()
+        return TopologyScorer([g.index for g in info.devices], bw)
+
+    # -- rpc handlers ------------------------------------------------------
+
+    def get_device_plugin_options(self, request, context):
+        return dpapi.DevicePluginOptions(
+            pre_start_required=False, get_preferred_allocation_available=True
+        )
+
+    def list_and_watch(self, request, context):
+        """Stream the device list; re-send on health-loop ticks."""
+        while not self._stop.is_set():
+            yield dpapi.ListAndWatchResponse(devices=self._device_list())
+            with self._update:
+                self._update.wait(timeout=self.health_interval_s)
+
+    def get_preferred_allocation(self, request, context):
+        """xGMI-best subset per container request."""
+        self.manager.update_gpu_info()
+        scorer = self._refresh_scorer()
+        responses = []
+        for creq in request.container_requests:
+            avail = list(creq.available_deviceIDs)
+            must = list(creq.must_include_deviceIDs)
+            k = creq.allocation_size
+            chosen: List[str]
+            if scorer is None or k >= len(avail):
+                chosen = avail[:k]
+            else:
+                uuid_to_idx = {
+                    u: self.manager.gpus[u].index
+                    for u in avail
+                    if u in self.manager.gpus
+                }
+                idx_to_uuid = {v: ku for ku, v in uuid_to_idx.items()}
+                free_idx = [uuid_to_idx[u] for u in avail if u in uuid_to_idx]
+                picked = scorer.choose(free_idx, k)
+                chosen = [idx_to_uuid[i] for i in picked]
+                # honour must_include (kubelet contract)
+                for m in must:
+                    if m not in chosen and chosen:
+                        chosen[-1] = m
+            responses.append(dpapi.ContainerPreferredAllocationResponse(deviceIDs=chosen))
+        return dpapi.PreferredAllocationResponse(container_responses=responses)
+
+    def allocate(self, request, context):
+        """Device IDs -> DeviceSpecs + env (SURVEY.md §3.3 analog)."""
+        out = []
+        for creq in request.container_requests:
+            specs = [
+                dpapi.DeviceSpec(
+                    container_path="/dev/kfd", host_path="/dev/kfd", permissions="rw"
+                )
+            ]
+            visible = []
+            for uuid in creq.devicesIDs:
+                g = self.manager.gpus.get(uuid)
+                if g is None:
+                    context.abort(
+                        grpc.StatusCode.INVALID_ARGUMENT, f"unknown device {uuid}"
+                    )
+                for path in (g.render_path, g.card_path):
+                    if path:
+                        specs.append(
+                            dpapi.DeviceSpec(
+                                container_path=path, host_path=path, permissions="rw"
+                            )
+                        )
+                visible.append(uuid)
+                g.in_use = True
+            resp = dpapi.ContainerAllocateResponse(devices=specs)
+            if visible:
+                resp.envs["ROCR_VISIBLE_DEVICES"] = ",".join(visible)
+            out.append(resp)
+        return dpapi.AllocateResponse(container_responses=out)
+
+    def pre_start_container(self, request, context):
+        return dpapi.PreStartContainerResponse()
+
+    def notify(self) -> None:
+        with self._update:
+            self._update.notify_all()
+
+    def stop(self) -> None:
+        self._stop.set()
+        self.notify()
+
+
+def _handlers(servicer: DevicePluginServicer) -> grpc.GenericRpcHandler:
+    rpcs = {
+        "GetDevicePluginOptions": grpc.unary_unary_rpc_method_handler(
+            servicer.get_device_plugin_options,
+            request_deserializer=dpapi.Empty.FromString,
+            response_serializer=_serialize,
+        ),
+        "ListAndWatch": grpc.unary_stream_rpc_method_handler(
+            servicer.list_and_watch,
+            request_deserializer=dpapi.Empty.FromString,
+            response_serializer=_serialize,
+        ),
+        "GetPreferredAllocation": grpc.unary_unary_rpc_method_handler(
+            servicer.get_preferred_allocation,
+            request_deserializer=dpapi.PreferredAllocationRequest.FromString,
+            response_serializer=_serialize,
+        ),
+        "Allocate": grpc.unary_unary_rpc_method_handler(
+            servicer.allocate,
+            request_deserializer=dpapi.AllocateRequest.FromString,
+            response_serializer=_serialize,
+        ),
+        "PreStartContainer": grpc.unary_unary_rpc_method_handler(
+            servicer.pre_start_container,
+            request_deserializer=dpapi.PreStartContainerRequest.FromString,
+            response_serializer=_serialize,
+        ),
+    }
+    return grpc.method_handlers_generic_handler(dpapi.DEVICE_PLUGIN_SERVICE, rpcs)
+
+
+class KubeletDevicePlugin:
+    """Lifecycle: serve on a unix socket + register with kubelet."""
+
+    def __init__(
+        self,
+        manager: AMDGPUManager,
+        socket_path: Optional[str] = None,
+        plugin_dir: str = dpapi.DEVICE_PLUGIN_PATH,
+        resource_name: str = RESOURCE_GPU,
+    ):
+        self.manager = manager
+        self.resource_name = resource_name
+        self.plugin_dir = plugin_dir
+        self.socket_path = socket_path or os.path.join(plugin_dir, "amdgpu.sock")
+        self.servicer = DevicePluginServicer(manager)
+        self._server: Optional[grpc.Server] = None
+
+    def start(self) -> str:
+        if os.path.exists(self.socket_path):
+            os.unlink(self.socket_path)
+        self._server = grpc.server(futures.ThreadPoolExecutor(max_workers=8))
+        self._server.add_generic_rpc_handlers((_handlers(self.servicer),))
+        self._server.add_insecure_port(f"unix://{self.socket_path}")
+        self._server.start()
+        utils.logf(1, "device plugin serving on %s", self.socket_path)
+        return self.socket_path
+
+    def register_with_kubelet(self, kubelet_socket: str = dpapi.KUBELET_SOCKET) -> None:
+        """POST our endpoint to kubelet's Registration service."""
+        channel = grpc.insecure_channel(f"unix://{kubelet_socket}")
+        register = channel.unary_unary(
+            f"/{dpapi.REGISTRATION_SERVICE}/Register",
+            request_serializer=_serialize,
+            response_deserializer=dpapi.Empty.FromString,
+        )
+        register(
+            dpapi.RegisterRequest(
+                version=dpapi.VERSION,
+                endpoint=os.path.basename(self.socket_path),
+                resource_name=self.resource_name,
+                options=dpapi.DevicePluginOptions(
+                    get_preferred_allocation_available=True
+                ),
+            )
+        )
+        utils.logf(1, "registered %s with kubelet", self.resource_name)
+
+    def stop(self) -> None:
+        self.servicer.stop()
+        if self._server is not None:
+            self._server.stop(grace=1.0)
+        if os.path.exists(self.socket_path):
+            try:
+                os.unlink(self.socket_path)
+            except OSError:
+                pass

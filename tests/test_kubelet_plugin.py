@@ -1,0 +1,164 @@
+"""Kubelet device-plugin server tests: real gRPC over a unix socket,
+fixture-backed manager (config 1 — the CPU-only 'kind cluster' analog)."""
+
+import os
+
+import grpc
+import pytest
+
+from kubegpu_amd.deviceplugin import create_device_plugin
+from kubegpu_amd.discovery import FakeBackend, fixtures
+from kubegpu_amd.server import KubeletDevicePlugin, dpapi
+
+
+def _serialize(m):
+    return m.SerializeToString()
+
+
+@pytest.fixture
+def plugin(tmp_path):
+    mgr = create_device_plugin(FakeBackend(fixtures.fixture_2hive_8gpu()))
+    mgr.start()
+    p = KubeletDevicePlugin(mgr, socket_path=str(tmp_path / "amdgpu.sock"))
+    p.start()
+    yield p
+    p.stop()
+
+
+@pytest.fixture
+def channel(plugin):
+    ch = grpc.insecure_channel(f"unix://{plugin.socket_path}")
+    yield ch
+    ch.close()
+
+
+def _stub(channel, method, req_cls, resp_cls, streaming=False):
+    path = f"/{dpapi.DEVICE_PLUGIN_SERVICE}/{method}"
+    if streaming:
+        return channel.unary_stream(path, request_serializer=_serialize,
+                                    response_deserializer=resp_cls.FromString)
+    return channel.unary_unary(path, request_serializer=_serialize,
+                               response_deserializer=resp_cls.FromString)
+
+
+def test_options(channel):
+    opts = _stub(channel, "GetDevicePluginOptions", dpapi.Empty,
+                 dpapi.DevicePluginOptions)(dpapi.Empty(), timeout=10)
+    assert opts.get_preferred_allocation_available
+
+
+def test_list_and_watch_first_frame(channel):
+    stream = _stub(channel, "ListAndWatch", dpapi.Empty,
+                   dpapi.ListAndWatchResponse, streaming=True)(
+        dpapi.Empty(), timeout=10
+    )
+    frame = next(iter(stream))
+    assert len(frame.devices) == 8
+    d = frame.devices[0]
+    assert d.health == dpapi.HEALTHY
+    assert d.ID.startswith("GPU-mi355x-")
+    assert len(d.topology.nodes) == 1
+
+
+def test_preferred_allocation_same_hive(channel):
+    """kubelet asks for 4 of 8 on a 2-hive node: must get one intact
+    hive, never a straddling set."""
+    req = dpapi.PreferredAllocationRequest(
+        container_requests=[
+            dpapi.ContainerPreferredAllocationRequest(
+                available_deviceIDs=[f"GPU-mi355x-{i:02d}" for i in range(8)],
+                allocation_size=4,
+            )
+        ]
+    )
+    resp = _stub(channel, "GetPreferredAllocation",
+                 dpapi.PreferredAllocationRequest,
+                 dpapi.PreferredAllocationResponse)(req, timeout=10)
+    ids = sorted(resp.container_responses[0].deviceIDs)
+    idx = [int(u.split("-")[-1]) for u in ids]
+    assert idx == [0, 1, 2, 3] or idx == [4, 5, 6, 7]
+
+
+def test_preferred_allocation_partial_free(channel):
+    """2 of {2,3,4,5}: pick the same-hive pair (2,3) or (4,5)."""
+    req = dpapi.PreferredAllocationRequest(
+        container_requests=[
+            dpapi.ContainerPreferredAllocationRequest(
+                available_deviceIDs=[f"GPU-mi355x-{i:02d}" for i in (2, 3, 4, 5)],
+                allocation_size=2,
+            )
+        ]
+    )
+    resp = _stub(channel, "GetPreferredAllocation",
+                 dpapi.PreferredAllocationRequest,
+                 dpapi.PreferredAllocationResponse)(req, timeout=10)
+    idx = sorted(int(u.split("-")[-1]) for u in resp.container_responses[0].deviceIDs)
+    assert idx in ([2, 3], [4, 5])
+
+
+def test_allocate_device_specs(channel, plugin):
+    req = dpapi.AllocateRequest(
+        container_requests=[
+            dpapi.ContainerAllocateRequest(
+                devicesIDs=["GPU-mi355x-01", "GPU-mi355x-02"]
+            )
+        ]
+    )
+    resp = _stub(channel, "Allocate", dpapi.AllocateRequest,
+                 dpapi.AllocateResponse)(req, timeout=10)
+    cr = resp.container_responses[0]
+    paths = [d.host_path for d in cr.devices]
+    assert "/dev/kfd" in paths
+    assert "/dev/dri/renderD129" in paths and "/dev/dri/renderD130" in paths
+    assert cr.envs["ROCR_VISIBLE_DEVICES"] == "GPU-mi355x-01,GPU-mi355x-02"
+    assert plugin.manager.gpus["GPU-mi355x-01"].in_use
+
+
+def test_allocate_unknown_device_rejected(channel):
+    req = dpapi.AllocateRequest(
+        container_requests=[
+            dpapi.ContainerAllocateRequest(devicesIDs=["GPU-bogus"])
+        ]
+    )
+    with pytest.raises(grpc.RpcError) as err:
+        _stub(channel, "Allocate", dpapi.AllocateRequest,
+              dpapi.AllocateResponse)(req, timeout=10)
+    assert err.value.code() == grpc.StatusCode.INVALID_ARGUMENT
+
+
+def test_registration_roundtrip(tmp_path):
+    """Register against a fake kubelet Registration service."""
+    from concurrent import futures
+
+    received = {}
+
+    def register(request, context):
+        received["req"] = request
+        return dpapi.Empty()
+
+    handler = grpc.method_handlers_generic_handler(
+        dpapi.REGISTRATION_SERVICE,
+        {
+            "Register": grpc.unary_unary_rpc_method_handler(
+                register,
+                request_deserializer=dpapi.RegisterRequest.FromString,
+                response_serializer=_serialize,
+            )
+        },
+    )
+    kubelet_sock = str(tmp_path / "kubelet.sock")
+    server = grpc.server(futures.ThreadPoolExecutor(max_workers=2))
+    server.add_generic_rpc_handlers((handler,))
+    server.add_insecure_port(f"unix://{kubelet_sock}")
+    server.start()
+
+    mgr = create_device_plugin(FakeBackend(fixtures.fixture_8x_mi355x()))
+    mgr.start()
+    p = KubeletDevicePlugin(mgr, socket_path=str(tmp_path / "amdgpu.sock"))
+    p.start()
+    p.register_with_kubelet(kubelet_socket=kubelet_sock)
+    assert received["req"].resource_name == "amd.com/gpu"
+    assert received["req"].version == "v1beta1"
+    assert received["req"].endpoint == "amdgpu.sock"
+    p.stop()
+    server.stop(grace=0.5)
