@@ -18,6 +18,10 @@
 
 #define WAVE 64
 #define BLOCK 256
+// Measured on MI355X (gpurun copy sweep, 1 GiB buffers): 1024 blocks
+// (4 per CU) + nontemporal beats larger grids — 5816 vs 4533 GB/s at
+// 4096 blocks; NT avoids LLC pollution on pure streams.
+#define DEFAULT_COPY_BLOCKS 1024
 
 __global__ void copy_kernel_v4(const uint4* __restrict__ src,
                                uint4* __restrict__ dst, size_t n4) {
@@ -87,9 +91,9 @@ void copy(at::Tensor dst, at::Tensor src) {
   auto stream = at::hip::getCurrentHIPStream();
   if (nbytes % 16 == 0) {
     size_t n4 = nbytes / 16;
-    int blocks = (int)std::min<size_t>((n4 + BLOCK - 1) / BLOCK, 4096);
-    hipLaunchKernelGGL(copy_kernel_v4, dim3(blocks), dim3(BLOCK), 0, stream,
-                       (const uint4*)src.data_ptr(), (uint4*)dst.data_ptr(), n4);
+    int blocks = (int)std::min<size_t>((n4 + BLOCK - 1) / BLOCK, DEFAULT_COPY_BLOCKS);
+    hipLaunchKernelGGL(copy_kernel_v4_nt, dim3(blocks), dim3(BLOCK), 0, stream,
+                       (const uint4v*)src.data_ptr(), (uint4v*)dst.data_ptr(), n4);
   } else {
     int blocks = (int)std::min<size_t>((nbytes + BLOCK - 1) / BLOCK, 4096);
     hipLaunchKernelGGL(copy_kernel_b, dim3(blocks), dim3(BLOCK), 0, stream,
@@ -113,7 +117,7 @@ double copy_bw_gbps(int64_t nbytes, int64_t iters, int64_t blocks_arg,
   size_t n4 = (size_t)nbytes / 16;
   int blocks = blocks_arg > 0
                    ? (int)blocks_arg
-                   : (int)std::min<size_t>((n4 + BLOCK - 1) / BLOCK, 4096);
+                   : (int)std::min<size_t>((n4 + BLOCK - 1) / BLOCK, DEFAULT_COPY_BLOCKS);
   auto launch = [&]() {
     if (nontemporal)
       hipLaunchKernelGGL(copy_kernel_v4_nt, dim3(blocks), dim3(BLOCK), 0, stream,
@@ -175,7 +179,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("copy", &copy, "streaming uint4 copy kernel (dst, src)");
   m.def("copy_bw_gbps", &copy_bw_gbps, "timed d2d copy bandwidth",
         py::arg("nbytes"), py::arg("iters") = 20, py::arg("blocks") = 0,
-        py::arg("nontemporal") = false);
+        py::arg("nontemporal") = true);
   m.def("read_bw_gbps", &read_bw_gbps, "timed read bandwidth",
         py::arg("nbytes"), py::arg("iters") = 20);
 }
