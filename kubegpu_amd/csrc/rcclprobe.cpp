@@ -26,7 +26,10 @@
 #include <hip/hip_bfloat16.h>
 #include <rccl/rccl.h>
 
+#include <hip/hip_fp16.h>
+
 #include <chrono>
+#include <cmath>
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
@@ -53,9 +56,23 @@
     }                                                                          \
   } while (0)
 
+__global__ void fill_bf16(unsigned short* p, size_t n, unsigned short bits) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) p[i] = bits;
+}
+
+static float bf16_to_float(unsigned short b) {
+  unsigned int u = (unsigned int)b << 16;
+  float f;
+  memcpy(&f, &u, sizeof(f));
+  return f;
+}
+
 int main(int argc, char** argv) {
   long long bytes = 256ll << 20;  // 256 MiB per rank
   int iters = 20, warmup = 5, ndev = -1;
+  bool check = true;
   std::string devlist;
   for (int i = 1; i < argc; ++i) {
     std::string a = argv[i];
@@ -71,6 +88,7 @@ int main(int argc, char** argv) {
     else if (a == "--warmup") warmup = atoi(next());
     else if (a == "--ndev") ndev = atoi(next());
     else if (a == "--devices") devlist = next();
+    else if (a == "--no-check") check = false;
     else {
       fprintf(stderr,
               "usage: rcclprobe [--ndev N | --devices 0,1,..] [--bytes B] "
@@ -104,7 +122,11 @@ int main(int argc, char** argv) {
     HIPCHECK(hipSetDevice(devs[i]));
     HIPCHECK(hipMalloc(&send[i], count * 2));
     HIPCHECK(hipMalloc(&recv[i], count * 2));
-    HIPCHECK(hipMemset(send[i], 0x3c, count * 2));  // bf16 pattern
+    // bf16 1.0 everywhere: after all-reduce(sum) every element == n,
+    // exactly representable -> bitwise-checkable numerics.
+    hipLaunchKernelGGL(fill_bf16, dim3(1024), dim3(256), 0, nullptr,
+                       (unsigned short*)send[i], count, (unsigned short)0x3F80);
+    HIPCHECK(hipDeviceSynchronize());
     HIPCHECK(hipStreamCreate(&streams[i]));
   }
 
@@ -133,6 +155,25 @@ int main(int argc, char** argv) {
   sync_all();
   auto t1 = std::chrono::steady_clock::now();
 
+  // numerics check: every element of every rank's recv == (float)n
+  // (one all-reduce of ones; iters>1 reduce the same send buffer, so the
+  // result is n after every iteration)
+  int bad = 0;
+  if (check) {
+    std::vector<unsigned short> host(256);
+    for (int i = 0; i < n && bad == 0; ++i) {
+      HIPCHECK(hipSetDevice(devs[i]));
+      HIPCHECK(hipMemcpy(host.data(), recv[i], host.size() * 2,
+                         hipMemcpyDeviceToHost));
+      for (unsigned short v : host)
+        if (std::fabs(bf16_to_float(v) - (float)n) > 1e-3f * n) ++bad;
+    }
+    if (bad) {
+      fprintf(stderr, "rcclprobe: NUMERICS CHECK FAILED (%d bad elements)\n", bad);
+      return 3;
+    }
+  }
+
   double sec = std::chrono::duration<double>(t1 - t0).count();
   double per_iter = sec / iters;
   double algbw = (double)(count * 2) / per_iter / 1e9;  // GB/s per rank payload
@@ -142,8 +183,9 @@ int main(int argc, char** argv) {
   printf(
       "{\"ndev\": %d, \"bytes\": %lld, \"iters\": %d, \"warmup\": %d, "
       "\"time_ms_per_iter\": %.3f, \"algbw_gbps\": %.2f, \"busbw_gbps\": %.2f, "
-      "\"dtype\": \"bf16\"}\n",
-      n, (long long)(count * 2), iters, warmup, per_iter * 1e3, algbw, busbw);
+      "\"dtype\": \"bf16\", \"check\": \"%s\"}\n",
+      n, (long long)(count * 2), iters, warmup, per_iter * 1e3, algbw, busbw,
+      check ? "pass" : "skipped");
 
   for (int i = 0; i < n; ++i) NCCLCHECK(ncclCommDestroy(comms[i]));
   for (int i = 0; i < n; ++i) {

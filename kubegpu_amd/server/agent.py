@@ -1,0 +1,99 @@
+"""Node agent — the long-running production entry point.
+
+Runs the device-plugin gRPC server, registers with kubelet (with retry —
+kubelet restarts wipe the plugin registry, so the agent watches its own
+socket and re-registers when the plugin dir is recreated), keeps
+discovery fresh, and serves Prometheus metrics.
+
+Usage:
+    python -m kubegpu_amd.server.agent [--socket PATH] [--plugin-dir DIR]
+        [--kubelet-socket PATH] [--no-register] [--metrics-port 9400]
+        [--fake]  (fixture backend, for plumbing tests without a GPU)
+"""
+
+from __future__ import annotations
+
+import argparse
+import os
+import signal
+import sys
+import time
+
+from ..api import utils
+from ..deviceplugin import create_device_plugin
+from ..discovery import FakeBackend, default_backend, fixtures
+from ..metrics import METRICS
+from . import dpapi
+from .kubelet_plugin import KubeletDevicePlugin
+
+
+def main(argv=None) -> int:
+    ap = argparse.ArgumentParser(prog="kubegpu-amd-agent")
+    ap.add_argument("--socket", default=None)
+    ap.add_argument("--plugin-dir", default=dpapi.DEVICE_PLUGIN_PATH)
+    ap.add_argument("--kubelet-socket", default=dpapi.KUBELET_SOCKET)
+    ap.add_argument("--no-register", action="store_true")
+    ap.add_argument("--metrics-port", type=int, default=9400)
+    ap.add_argument("--health-interval", type=float, default=30.0)
+    ap.add_argument("--fake", action="store_true",
+                    help="use the 8xMI355X fixture backend (no GPU needed)")
+    ap.add_argument("--oneshot", action="store_true",
+                    help="start, print state, exit (plumbing check)")
+    args = ap.parse_args(argv)
+
+    backend = (
+        FakeBackend(fixtures.fixture_8x_mi355x()) if args.fake else default_backend()
+    )
+    manager = create_device_plugin(backend)
+    manager.start()
+    utils.logf(0, "agent: discovered %d GPU(s)", len(manager.gpus))
+
+    plugin = KubeletDevicePlugin(
+        manager,
+        socket_path=args.socket,
+        plugin_dir=args.plugin_dir,
+    )
+    plugin.servicer.health_interval_s = args.health_interval
+    plugin.start()
+
+    if args.metrics_port:
+        if METRICS.serve(args.metrics_port):
+            utils.logf(1, "agent: metrics on :%d", args.metrics_port)
+
+    stop = {"flag": False}
+
+    def _sig(_signo, _frame):
+        stop["flag"] = True
+
+    signal.signal(signal.SIGTERM, _sig)
+    signal.signal(signal.SIGINT, _sig)
+
+    registered = False
+    if args.oneshot:
+        print(f"agent ok: {len(manager.gpus)} GPUs, socket {plugin.socket_path}")
+        plugin.stop()
+        return 0
+
+    while not stop["flag"]:
+        if not args.no_register and not registered:
+            try:
+                plugin.register_with_kubelet(args.kubelet_socket)
+                registered = True
+            except Exception as e:
+                utils.logf(2, "agent: kubelet registration pending: %s", e)
+        # kubelet restart detection: our socket vanishes when the plugin
+        # dir is recreated -> re-serve + re-register
+        if not os.path.exists(plugin.socket_path):
+            utils.logf(0, "agent: socket vanished (kubelet restart?); re-serving")
+            plugin.stop()
+            plugin.start()
+            registered = False
+        plugin.servicer.notify()  # wake ListAndWatch to refresh health
+        time.sleep(args.health_interval if registered else 5.0)
+
+    plugin.stop()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

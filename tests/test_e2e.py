@@ -158,3 +158,32 @@ def test_scheduler_latency_sanity():
         cluster.release(pod)
     lat.sort()
     assert lat[len(lat) // 2] < 0.25  # generous CI bound
+
+
+def test_flat_no_topology_path():
+    """gpu-generate-topology=0: wildcard requests bind through the core
+    and still land on the xGMI-best subset."""
+    from kubegpu_amd.scheduler import GPU_TOPOLOGY_GENERATION
+
+    cluster = _cluster_with(("twohive", fixtures.fixture_2hive_8gpu()))
+    pod = PodInfo(
+        name="flat",
+        requests={GPU_TOPOLOGY_GENERATION: 0},
+        running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: 2})},
+    )
+    res = cluster.schedule(pod)
+    assert len(res.uuids) == 2
+    idx = sorted(int(u.split("-")[-1]) for u in res.uuids)
+    assert idx[0] // 4 == idx[1] // 4  # same hive even via the flat path
+    reqs = list(pod.running_containers["c"].dev_requests)
+    assert all("/gpugrp1/*/gpugrp0/*/" in r for r in reqs)
+
+
+def test_agent_oneshot_runs():
+    from kubegpu_amd.server import agent
+
+    rc = agent.main([
+        "--fake", "--no-register", "--metrics-port", "0",
+        "--socket", "/tmp/kubegpu-agent-test.sock", "--oneshot",
+    ])
+    assert rc == 0
