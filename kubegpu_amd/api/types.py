@@ -18,7 +18,6 @@ gpuschedulerplugin/gpu_test.go:79-84).
 
 from __future__ import annotations
 
-import threading
 from dataclasses import dataclass, field
 from typing import Dict, Optional
 
@@ -129,18 +128,3 @@ class Mount:
     container_path: str
     read_only: bool = True
 
-
-class Volatile:
-    """Small helper: a lock-guarded value holder used by managers/caches."""
-
-    def __init__(self, value=None):
-        self._lock = threading.Lock()
-        self._value = value
-
-    def get(self):
-        with self._lock:
-            return self._value
-
-    def set(self, value):
-        with self._lock:
-            self._value = value

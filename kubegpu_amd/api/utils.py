@@ -7,7 +7,6 @@ Levels follow the reference's usage: 0 = errors/always, 3-5 = debug detail.
 
 from __future__ import annotations
 
-import io
 import logging
 import os
 import sys
@@ -53,7 +52,3 @@ def sorted_string_keys(m: Mapping[str, object]) -> List[str]:
     """utils.SortedStringKeys: deterministic iteration order over maps."""
     return sorted(m.keys())
 
-
-def writer_for_log() -> io.StringIO:
-    """A string buffer for tree pretty-printing (used by plugintypes)."""
-    return io.StringIO()

@@ -36,13 +36,7 @@ from ..discovery import (
 )
 from ..scheduler.translate import SchedulingError
 from ..scheduler.treecache import LabeledLayout, parse_node_resources
-from ..scheduler.xgmi import BwMatrix, TopologyScorer, best_ring
-
-
-def _ring_of(indices, bw):
-    """best_ring with the k=1 infinity capped for safe score arithmetic."""
-    val, order = best_ring(indices, bw)
-    return (min(val, 1e9), order)
+from ..scheduler.xgmi import BwMatrix, TopologyScorer
 
 
 @dataclass
