@@ -74,7 +74,7 @@ def main() -> int:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--bytes", type=int, default=256 << 20)
+    ap.add_argument("--bytes", type=int, default=1 << 30)
     ap.add_argument("--pods", type=int, default=2000)
     args = ap.parse_args()
 

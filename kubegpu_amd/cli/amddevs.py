@@ -67,11 +67,13 @@ def main(argv=None) -> int:
     }, indent=1))
 
     if args.probe:
-        from ..probe import run_rccl_probe
+        from ..probe import probe_with_link_utilization, run_rccl_probe
 
         idxs = sorted(mgr.gpus[u].index for u in res.uuids)
-        out = run_rccl_probe(devices=idxs, nbytes=args.bytes)
-        print(json.dumps(out, indent=1))
+        out, links = probe_with_link_utilization(
+            run_rccl_probe, devices=idxs, nbytes=args.bytes
+        )
+        print(json.dumps({"probe": out, "xgmi_link_traffic": links}, indent=1))
     return 0
 
 

@@ -44,9 +44,13 @@ def _concretize_wildcards(resources: Dict[str, int]) -> None:
 
 
 class AMDGPUScheduler(DeviceScheduler):
-    def __init__(self, cache: Optional[NodeTreeCache] = None):
+    def __init__(self, cache: Optional[NodeTreeCache] = None, group_core=None):
         self.cache = cache if cache is not None else NodeTreeCache()
         self.topologies: Dict[str, GpusInfo] = {}
+        # optional kubegpu_amd.core.GroupScheduler: lets pod_fits_device
+        # run the concrete binder when run_group_scheduler=True (the
+        # reference leaves this to the external core; we own it)
+        self.group_core = group_core
 
     # -- node lifecycle ----------------------------------------------------
 
@@ -79,6 +83,9 @@ class AMDGPUScheduler(DeviceScheduler):
         target = pod_info if fill_allocate_from else pod_info.copy()
         try:
             translate_pod_gpu_resources(node_info, target, self.cache)
+            if run_group_scheduler and self.group_core is not None:
+                # dry-run concrete binding on this node (commit=False)
+                self.group_core.bind_pod(node_info.name, target.copy(), commit=False)
         except SchedulingError as e:
             utils.logf(3, "pod %s does not fit: %s", pod_info.name, e)
             reason = PredicateFailureReason(

@@ -187,3 +187,20 @@ def test_agent_oneshot_runs():
         "--socket", "/tmp/kubegpu-agent-test.sock", "--oneshot",
     ])
     assert rc == 0
+
+
+def test_pod_fits_with_group_scheduler_dry_run():
+    """pod_fits_device(run_group_scheduler=True) exercises the concrete
+    binder: a full node stops fitting even though translation exists."""
+    cluster = _cluster_with(("dense", fixtures.fixture_8x_mi355x()))
+    cluster.scheduler.group_core = cluster.core
+    ni = cluster.node_infos["dense"]
+    fits, _, _ = cluster.scheduler.pod_fits_device(
+        ni, _pod("f2", 2), fill_allocate_from=False, run_group_scheduler=True
+    )
+    assert fits
+    cluster.schedule(_pod("all8", 8))  # fill the node
+    fits, reasons, _ = cluster.scheduler.pod_fits_device(
+        ni, _pod("f2b", 2), fill_allocate_from=False, run_group_scheduler=True
+    )
+    assert not fits and reasons
