@@ -45,6 +45,43 @@ __global__ void copy_kernel_v4_nt(const uint4v* __restrict__ src,
   }
 }
 
+// 4x unrolled NT variant: four independent uint4 transfers in flight
+// per thread per iteration (more memory-level parallelism per wave).
+__global__ void copy_kernel_v4_nt_u4(const uint4v* __restrict__ src,
+                                     uint4v* __restrict__ dst, size_t n4) {
+  size_t tid = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  size_t i = tid;
+  size_t bound = n4 >= 3 * stride ? n4 - 3 * stride : 0;
+  for (; i < bound; i += 4 * stride) {
+    uint4v a = __builtin_nontemporal_load(&src[i]);
+    uint4v b = __builtin_nontemporal_load(&src[i + stride]);
+    uint4v c = __builtin_nontemporal_load(&src[i + 2 * stride]);
+    uint4v d = __builtin_nontemporal_load(&src[i + 3 * stride]);
+    __builtin_nontemporal_store(a, &dst[i]);
+    __builtin_nontemporal_store(b, &dst[i + stride]);
+    __builtin_nontemporal_store(c, &dst[i + 2 * stride]);
+    __builtin_nontemporal_store(d, &dst[i + 3 * stride]);
+  }
+  for (; i < n4; i += stride) {
+    uint4v v = __builtin_nontemporal_load(&src[i]);
+    __builtin_nontemporal_store(v, &dst[i]);
+  }
+}
+
+// Contiguous-chunk partitioning: block b owns one contiguous region
+// (DRAM-page-friendly; each wave still issues coalesced 1 KiB lines).
+__global__ void copy_kernel_v4_nt_chunk(const uint4v* __restrict__ src,
+                                        uint4v* __restrict__ dst, size_t n4) {
+  size_t per_block = (n4 + gridDim.x - 1) / gridDim.x;
+  size_t begin = (size_t)blockIdx.x * per_block;
+  size_t end = begin + per_block < n4 ? begin + per_block : n4;
+  for (size_t i = begin + threadIdx.x; i < end; i += blockDim.x) {
+    uint4v v = __builtin_nontemporal_load(&src[i]);
+    __builtin_nontemporal_store(v, &dst[i]);
+  }
+}
+
 __global__ void copy_kernel_b(const unsigned char* __restrict__ src,
                               unsigned char* __restrict__ dst, size_t n) {
   size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -107,7 +144,7 @@ void copy(at::Tensor dst, at::Tensor src) {
 // (bytes read + bytes written over wall time, hipEvent-timed).
 // blocks=0 picks the default; nontemporal selects the NT variant.
 double copy_bw_gbps(int64_t nbytes, int64_t iters, int64_t blocks_arg,
-                    bool nontemporal) {
+                    bool nontemporal, int64_t variant) {
   TORCH_CHECK(nbytes > 0 && nbytes % 16 == 0, "nbytes must be positive, 16-aligned");
   auto opts = at::TensorOptions().dtype(at::kByte).device(at::kCUDA);
   at::Tensor src = at::empty({nbytes}, opts);
@@ -119,7 +156,14 @@ double copy_bw_gbps(int64_t nbytes, int64_t iters, int64_t blocks_arg,
                    ? (int)blocks_arg
                    : (int)std::min<size_t>((n4 + BLOCK - 1) / BLOCK, DEFAULT_COPY_BLOCKS);
   auto launch = [&]() {
-    if (nontemporal)
+    if (variant == 1)
+      hipLaunchKernelGGL(copy_kernel_v4_nt_u4, dim3(blocks), dim3(BLOCK), 0, stream,
+                         (const uint4v*)src.data_ptr(), (uint4v*)dst.data_ptr(), n4);
+    else if (variant == 2)
+      hipLaunchKernelGGL(copy_kernel_v4_nt_chunk, dim3(blocks), dim3(BLOCK), 0,
+                         stream, (const uint4v*)src.data_ptr(),
+                         (uint4v*)dst.data_ptr(), n4);
+    else if (nontemporal)
       hipLaunchKernelGGL(copy_kernel_v4_nt, dim3(blocks), dim3(BLOCK), 0, stream,
                          (const uint4v*)src.data_ptr(), (uint4v*)dst.data_ptr(), n4);
     else
@@ -139,6 +183,8 @@ double copy_bw_gbps(int64_t nbytes, int64_t iters, int64_t blocks_arg,
   (void)hipEventDestroy(t0);
   (void)hipEventDestroy(t1);
   C10_HIP_KERNEL_LAUNCH_CHECK();
+  TORCH_CHECK(at::equal(dst, src), "copy_bw_gbps: variant ", variant,
+              " produced wrong output");
   double sec = ms / 1e3;
   return (double)nbytes * 2.0 * iters / sec / 1e9;
 }
@@ -179,7 +225,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("copy", &copy, "streaming uint4 copy kernel (dst, src)");
   m.def("copy_bw_gbps", &copy_bw_gbps, "timed d2d copy bandwidth",
         py::arg("nbytes"), py::arg("iters") = 20, py::arg("blocks") = 0,
-        py::arg("nontemporal") = true);
+        py::arg("nontemporal") = true, py::arg("variant") = 0);
   m.def("read_bw_gbps", &read_bw_gbps, "timed read bandwidth",
         py::arg("nbytes"), py::arg("iters") = 20);
 }

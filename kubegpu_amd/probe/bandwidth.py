@@ -55,10 +55,14 @@ def d2d_copy_bw_gbps(
     nbytes: int = 1 << 30,
     iters: int = 20,
     blocks: int = 0,
-    nontemporal: bool = False,
+    nontemporal: bool = True,
+    variant: int = 0,
 ) -> float:
-    """Timed device-to-device streaming-copy bandwidth (GB/s, R+W)."""
-    return load_ext().copy_bw_gbps(nbytes, iters, blocks, nontemporal)
+    """Timed device-to-device streaming-copy bandwidth (GB/s, R+W).
+
+    variant: 0 = grid-stride, 1 = 4x-unrolled, 2 = contiguous-chunk.
+    """
+    return load_ext().copy_bw_gbps(nbytes, iters, blocks, nontemporal, variant)
 
 
 def read_bw_gbps(nbytes: int = 1 << 30, iters: int = 20) -> float:
