@@ -83,6 +83,14 @@ def main() -> int:
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    if args.gpus > 1 and world == 1:
+        print(
+            f"bench.py: --gpus {args.gpus} needs one rank per GPU; launch via "
+            f"python -m torch.distributed.run --nnodes=1 --nproc-per-node "
+            f"{args.gpus} --master-addr 127.0.0.1 bench.py --gpus {args.gpus} ...",
+            file=sys.stderr,
+        )
+        return 2
     n_gpus = world if world > 1 else args.gpus
     on_gpu = torch.cuda.is_available()
 

@@ -76,3 +76,27 @@ def test_torch_allreduce_probe_world2(tmp_path):
     res = _run_dist(str(script))
     assert res.returncode == 0, res.stderr[-2000:]
     assert "PROBE_OK" in res.stdout
+
+
+@pytest.mark.timeout(180)
+def test_bench_single_process_contract():
+    """Default (N=1) bench prints one valid contract line on CPU."""
+    res = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"),
+         "--steps", "2", "--warmup", "1", "--pods", "20"],
+        capture_output=True, timeout=150, cwd=REPO, text=True,
+    )
+    assert res.returncode == 0, res.stderr[-2000:]
+    rec = json.loads(res.stdout.strip().splitlines()[-1])
+    assert rec["n_gpus"] == 1 and rec["value"] > 0
+    assert rec["metric"] == "scheduled_set_allreduce_busbw_GBps"
+
+
+def test_bench_gpus_flag_without_ranks_errors():
+    res = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"), "--gpus", "4",
+         "--steps", "1", "--warmup", "0", "--pods", "5"],
+        capture_output=True, timeout=150, cwd=REPO, text=True,
+    )
+    assert res.returncode == 2
+    assert "torch.distributed.run" in res.stderr
