@@ -1,0 +1,90 @@
+"""Property-based soak tests (hypothesis): random pod streams against
+random topologies must preserve allocation invariants."""
+
+import hypothesis.strategies as st
+from hypothesis import given, settings
+
+from kubegpu_amd.api.types import ContainerInfo, PodInfo
+from kubegpu_amd.core import Cluster
+from kubegpu_amd.deviceplugin import create_device_plugin
+from kubegpu_amd.discovery import FakeBackend, fixtures
+from kubegpu_amd.plugintypes import RESOURCE_GPU
+from kubegpu_amd.scheduler import SchedulingError
+
+FIXTURES = {
+    "dense": fixtures.fixture_8x_mi355x,
+    "twohive": fixtures.fixture_2hive_8gpu,
+    "noxgmi": fixtures.fixture_4x_no_xgmi,
+}
+
+
+def _mk_cluster(names):
+    cluster = Cluster()
+    for i, n in enumerate(names):
+        mgr = create_device_plugin(FakeBackend(FIXTURES[n]()))
+        cluster.add_node_from_manager(f"{n}-{i}", mgr)
+    return cluster
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    nodes=st.lists(st.sampled_from(sorted(FIXTURES)), min_size=1, max_size=3),
+    stream=st.lists(
+        st.tuples(st.integers(min_value=1, max_value=8), st.booleans()),
+        min_size=1,
+        max_size=30,
+    ),
+)
+def test_allocation_invariants(nodes, stream):
+    """For any node mix and any schedule/release stream:
+    - a pod's GPUs are distinct and all on its node
+    - no GPU is held by two live pods
+    - releasing everything restores full capacity
+    """
+    cluster = _mk_cluster(nodes)
+    total = {name: cluster.core.free_count(name) for name in cluster.core.nodes}
+    live = []
+    # fixtures reuse uuids across nodes, so the unique key is (node, uuid)
+    held = {}  # (node, uuid) -> pod name
+    for i, (k, do_release) in enumerate(stream):
+        pod = PodInfo(
+            name=f"p{i}",
+            running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: k})},
+        )
+        try:
+            res = cluster.schedule(pod)
+        except SchedulingError:
+            res = None
+        if res is not None:
+            assert len(set(res.uuids)) == len(res.uuids) == k
+            state = cluster.core.nodes[res.node_name]
+            for u in res.uuids:
+                assert u in state.gpus, "allocated GPU not on chosen node"
+                key = (res.node_name, u)
+                assert key not in held, f"double allocation of {key}"
+                held[key] = pod.name
+            live.append((pod, res))
+        if do_release and live:
+            pod0, res0 = live.pop(0)
+            cluster.release(pod0)
+            for u in res0.uuids:
+                del held[(res0.node_name, u)]
+    for pod0, res0 in live:
+        cluster.release(pod0)
+    for name, cap in total.items():
+        assert cluster.core.free_count(name) == cap
+
+
+@settings(max_examples=30, deadline=None)
+@given(k=st.integers(min_value=1, max_value=4))
+def test_same_hive_whenever_possible(k):
+    """On the 2-hive node, any k<=4 pod scheduled on an empty node must
+    stay inside one hive (the xGMI-optimal answer is always feasible)."""
+    cluster = _mk_cluster(["twohive"])
+    pod = PodInfo(
+        name="p",
+        running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: k})},
+    )
+    res = cluster.schedule(pod)
+    hives = {int(u.split("-")[-1]) // 4 for u in res.uuids}
+    assert len(hives) == 1
