@@ -87,6 +87,7 @@ def torch_allreduce_busbw(
         dist.all_reduce(buf)
     _sync()
     dist.barrier()
+    _sync()  # the barrier itself is a device op under NCCL
     t1 = time.perf_counter()
 
     elapsed = torch.tensor([t1 - t0], dtype=torch.float64)

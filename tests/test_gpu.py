@@ -134,3 +134,46 @@ def test_graft_smoke():
     import __graft_entry__ as g
 
     g.smoke()
+
+
+def test_probe_with_link_utilization_real():
+    """Counter-bracketed probe on the real box: structure is returned
+    even when the 1-GPU probe moves no xGMI traffic."""
+    from kubegpu_amd.probe import probe_with_link_utilization, run_rccl_probe
+
+    out, links = probe_with_link_utilization(
+        run_rccl_probe, ndev=1, nbytes=64 << 20, iters=3, warmup=1
+    )
+    assert out["busbw_gbps"] > 0
+    assert out.get("check") == "pass"
+    if links is not None:  # amdsmi counters available
+        assert 0 in links
+
+
+def test_amddevs_cli_plugin_mode():
+    import json as _json
+    import subprocess as _sp
+    import sys as _sys
+
+    res = _sp.run(
+        [_sys.executable, "-m", "kubegpu_amd.cli.amddevs", "--plugin"],
+        capture_output=True, timeout=120, cwd=REPO, text=True,
+    )
+    assert res.returncode == 0, res.stderr[-1000:]
+    rec = _json.loads(res.stdout)
+    assert rec["kube_alloc"]["amd.com/gpu"] >= 1
+
+
+def test_amddevs_cli_schedule_mode():
+    import json as _json
+    import subprocess as _sp
+    import sys as _sys
+
+    res = _sp.run(
+        [_sys.executable, "-m", "kubegpu_amd.cli.amddevs", "--schedule", "1"],
+        capture_output=True, timeout=120, cwd=REPO, text=True,
+    )
+    assert res.returncode == 0, res.stderr[-1000:]
+    rec = _json.loads(res.stdout)
+    assert len(rec["gpus"]) == 1
+    assert "/dev/kfd" in rec["devices"]
