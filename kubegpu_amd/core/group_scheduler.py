@@ -124,6 +124,13 @@ class GroupScheduler:
             state.bw = _synthetic_bw(list(state.gpus.values()))
         state.scorer = TopologyScorer(list(state.index_to_uuid.keys()), state.bw)
         with self._lock:
+            # node re-registration (watch update / re-discovery) must not
+            # forget live allocations: carry over used flags for GPUs
+            # that still exist (cf. the manager's in_use-survives-
+            # rediscovery rule, nvidia_gpu_manager.go:143-145)
+            prev = self.nodes.get(node_info.name)
+            if prev is not None:
+                state.used = {u for u in prev.used if u in state.gpus}
             self.nodes[node_info.name] = state
         return state
 

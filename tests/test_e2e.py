@@ -204,3 +204,38 @@ def test_pod_fits_with_group_scheduler_dry_run():
         ni, _pod("f2b", 2), fill_allocate_from=False, run_group_scheduler=True
     )
     assert not fits and reasons
+
+
+def test_node_readvertise_preserves_allocations():
+    """Re-adding a node (watch update / re-discovery) keeps live pods'
+    GPUs allocated; a vanished GPU drops out cleanly."""
+    from kubegpu_amd.api.types import NodeInfo
+    from kubegpu_amd.discovery import GpusInfo
+
+    cluster = _cluster_with(("dense", fixtures.fixture_8x_mi355x()))
+    pod = _pod("live", 2)
+    res = cluster.schedule(pod)
+    assert cluster.core.free_count("dense") == 6
+
+    # re-advertise the same node (e.g. periodic UpdateNodeInfo)
+    mgr = cluster.managers["dense"]
+    ni = NodeInfo(name="dense")
+    mgr.update_node_info(ni)
+    cluster.scheduler.add_node("dense", ni, mgr._last_info)
+    cluster.core.register_node(ni, mgr._last_info)
+    cluster.node_infos["dense"] = ni
+    assert cluster.core.free_count("dense") == 6  # allocations survive
+
+    # shrink the node: one held GPU vanishes
+    smaller = GpusInfo.from_json(mgr._last_info.to_json())
+    held = sorted(res.uuids)[0]
+    smaller.devices = [d for d in smaller.devices if d.uuid != held]
+    from kubegpu_amd.discovery import FakeBackend
+    mgr2 = create_device_plugin(FakeBackend(smaller))
+    ni2 = NodeInfo(name="dense")
+    mgr2.start()
+    mgr2.update_node_info(ni2)
+    cluster.core.register_node(ni2, smaller)
+    st = cluster.core.nodes["dense"]
+    assert held not in st.gpus
+    assert len(st.used) == 1  # the other held GPU is still tracked
