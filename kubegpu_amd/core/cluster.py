@@ -39,9 +39,11 @@ class ScheduleResult:
 class Cluster:
     """In-process cluster: device scheduler + group core + node plugins."""
 
-    def __init__(self, scheduler: Optional[AMDGPUScheduler] = None):
+    def __init__(self, scheduler: Optional[AMDGPUScheduler] = None,
+                 policy: str = "xgmi"):
         self.scheduler = scheduler or AMDGPUScheduler()
-        self.core = GroupScheduler()
+        self.core = GroupScheduler(policy=policy)
+        self.policy = policy
         self.node_infos: Dict[str, NodeInfo] = {}
         self.managers: Dict[str, AMDGPUManager] = {}
 
@@ -102,7 +104,13 @@ class Cluster:
             candidates.append((score, name, trial, uuids))
         if not candidates:
             raise SchedulingError(f"no node fits pod {pod.name}")
-        candidates.sort(key=lambda c: (c[0], c[1]), reverse=True)
+        if self.policy == "naive":
+            # reference-like: no bandwidth model across nodes — first
+            # fitting node in name order (the external core's arbitrary
+            # choice; fit returned score 0.0, gpu_scheduler.go:43)
+            candidates.sort(key=lambda c: c[1])
+        else:
+            candidates.sort(key=lambda c: (c[0], c[1]), reverse=True)
         _, node_name, bound_pod, uuids = candidates[0]
         # adopt the winning translation/bindings into the caller's pod
         pod.running_containers = bound_pod.running_containers

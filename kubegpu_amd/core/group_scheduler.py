@@ -83,11 +83,25 @@ def _synthetic_bw(gpus: List[_Gpu]) -> BwMatrix:
 
 
 class GroupScheduler:
-    """Concrete per-node request binder with xGMI-aware subset choice."""
+    """Concrete per-node request binder with xGMI-aware subset choice.
 
-    def __init__(self) -> None:
+    policy="xgmi" (default) scores subsets by ring bandwidth +
+    anti-fragmentation; policy="naive" reproduces the reference's
+    behavior (densest-group packing order but arbitrary — lowest-index —
+    GPU choice inside a group, first-fit group matching, no bandwidth
+    model), used by the policy-comparison bench (tools/compare_policies).
+    """
+
+    def __init__(self, policy: str = "xgmi") -> None:
         self._lock = threading.RLock()
         self.nodes: Dict[str, NodeState] = {}
+        assert policy in ("xgmi", "naive")
+        self.policy = policy
+
+    def _choose(self, state: NodeState, cand: List[int], k: int) -> List[int]:
+        if self.policy == "naive":
+            return sorted(cand)[:k] if k <= len(cand) else []
+        return state.scorer.choose(cand, k)
 
     # -- node registration -------------------------------------------------
 
@@ -195,7 +209,7 @@ class GroupScheduler:
                     members = self._group_members(state, ah, ag)
                     cand = [state.gpus[u].index for u in members if u in free]
                     k = len(demands[(hi, gi)])
-                    picked = state.scorer.choose(cand, k)
+                    picked = self._choose(state, cand, k)
                     picked_uuids = [state.index_to_uuid[i] for i in picked]
                     for (cont, req), uuid in zip(demands[(hi, gi)], picked_uuids):
                         bindings.append((cont, req, state.gpus[uuid].concrete_name))
@@ -207,7 +221,7 @@ class GroupScheduler:
             if wkey in demands:
                 k = len(demands[wkey])
                 cand = [state.gpus[u].index for u in sorted(free)]
-                picked = state.scorer.choose(cand, k)
+                picked = self._choose(state, cand, k)
                 if len(picked) < k:
                     raise SchedulingError(
                         f"node {node_name}: {len(cand)} free GPUs, pod "
@@ -267,6 +281,7 @@ class GroupScheduler:
                 free_per[(ah, ag)] = [u for u in ids if u in free]
 
         h_list = sorted(his, key=lambda h: -sum(his[h].values()))
+        naive = self.policy == "naive"
         best: List = [None, None]  # score, assignment
 
         def match_g(ah: int, gi_counts: List[Tuple[int, int]]):
@@ -288,13 +303,16 @@ class GroupScheduler:
                 n_free, ag = cands[0]
                 used_ag.add(ag)
                 out[gi] = ag
-                idxs = [state.gpus[u].index for u in free_per[(ah, ag)]]
-                picked = state.scorer.choose(idxs, count)
-                quality += state.scorer.ring_bw(picked)
+                if not naive:
+                    idxs = [state.gpus[u].index for u in free_per[(ah, ag)]]
+                    picked = state.scorer.choose(idxs, count)
+                    quality += state.scorer.ring_bw(picked)
                 leftover += n_free - count
             return out, (quality, -leftover)
 
         def rec(i: int, used_ah: Set[int], assign: Dict, score_acc: Tuple[float, int]):
+            if naive and best[1] is not None:
+                return  # first-fit: stop at the first feasible assignment
             if i == len(h_list):
                 if best[0] is None or score_acc > best[0]:
                     best[0], best[1] = score_acc, dict(assign)

@@ -115,6 +115,25 @@ def fixture_2hive_8gpu() -> GpusInfo:
     return GpusInfo(version=_version(), devices=gpus)
 
 
+def fixture_degraded_mesh(missing=((0, 1), (0, 2), (0, 3), (5, 6))) -> GpusInfo:
+    """8-GPU mesh with some xGMI links DOWN (pairs fall back to PCIe).
+
+    Models a production node with failed links (amdsmi exposes
+    xgmi_link_status up/down); the regime where per-link-aware subset
+    choice beats group-shape policies.
+    """
+    gpus = [_mk_gpu(i, numa=i // 4) for i in range(8)]
+    down = {(min(a, b), max(a, b)) for a, b in missing}
+    pairs = {
+        (i, j)
+        for i in range(8)
+        for j in range(i + 1, 8)
+        if (i, j) not in down
+    }
+    _connect(gpus, pairs)
+    return GpusInfo(version=_version(), devices=gpus)
+
+
 def fixture_4x_no_xgmi() -> GpusInfo:
     """Degenerate 4-GPU node without any xGMI (PCIe only)."""
     gpus = [_mk_gpu(i, numa=0) for i in range(4)]

@@ -101,3 +101,22 @@ def test_fast_path_matches_python(fixture_2hive):
         ([1, 3, 5, 7], 3),
     ]:
         assert choose_best_subset_fast(free, k, bw) == choose_best_subset(free, k, bw)
+
+
+def test_degraded_mesh_avoids_down_links():
+    """On a mesh with GPU0's links 0-1/0-2/0-3 down, a 4-GPU subset must
+    avoid routing through GPU0's dead links (naive [0,1,2,3] would be
+    PCIe-bound)."""
+    fix = fixtures.fixture_degraded_mesh(missing=((0, 1), (0, 2), (0, 3), (5, 6)))
+    bw = _bw(fix)
+    val, _ = best_ring([0, 1, 2, 3], bw)
+    assert val == 63.0  # the naive choice is PCIe-bound
+    picked = choose_best_subset(list(range(8)), 4, bw)
+    val2, order = best_ring(picked, bw)
+    assert val2 == 153.0
+    # every consecutive ring hop is a live xGMI link (the subset may
+    # contain endpoints of a dead link as long as the ring avoids it)
+    down = {(0, 1), (0, 2), (0, 3), (5, 6)}
+    hops = {tuple(sorted((order[i], order[(i + 1) % len(order)])))
+            for i in range(len(order))}
+    assert not (hops & down)
