@@ -43,3 +43,26 @@ class Device(abc.ABC):
     ) -> Tuple[List[Mount], List[str], Dict[str, str]]:
         """Resolve container.allocate_from into concrete (mounts, device
         paths, env vars) at container-create time."""
+
+
+def create_device_from_plugin(path: str) -> Device:
+    """Load a device plugin from a Python file path and instantiate it.
+
+    Parity with device.CreateDeviceFromPlugin loading a Go buildmode
+    plugin .so (reference cmd/main.go:23): the module must expose
+    ``create_device_plugin() -> Device``.
+    """
+    import importlib.util
+
+    spec = importlib.util.spec_from_file_location("kubegpu_amd_plugin", path)
+    if spec is None or spec.loader is None:
+        raise ImportError(f"cannot load device plugin from {path}")
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    factory = getattr(mod, "create_device_plugin", None)
+    if factory is None:
+        raise AttributeError(f"{path} does not export create_device_plugin()")
+    dev = factory()
+    if not isinstance(dev, Device):
+        raise TypeError(f"{path}: create_device_plugin() returned {type(dev)}")
+    return dev

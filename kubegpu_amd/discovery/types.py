@@ -200,6 +200,10 @@ class GpusInfo:
                         if l.type == LINK_XGMI
                         else PCIE_GBPS_DEFAULT
                     )
+                if l.type == LINK_XGMI and not l.p2p:
+                    # link advertised but peer access disabled/down:
+                    # traffic bounces through the host path
+                    gbps = min(gbps, PCIE_GBPS_DEFAULT)
                 bw[g.index][l.peer_index] = gbps
         for i in idx:
             for j in idx:

@@ -146,3 +146,32 @@ def test_discovery_cache(fixture_8x):
     assert calls["n"] == 1
     mgr.update_gpu_info(force=True)
     assert calls["n"] == 2
+
+
+def test_create_device_from_plugin(tmp_path):
+    """Plugin loading parity (device.CreateDeviceFromPlugin analog)."""
+    from kubegpu_amd.api.device import create_device_from_plugin
+
+    plugin_file = tmp_path / "myplugin.py"
+    plugin_file.write_text(
+        "from kubegpu_amd.deviceplugin import create_device_plugin as _f\n"
+        "from kubegpu_amd.discovery import FakeBackend, fixtures\n"
+        "def create_device_plugin():\n"
+        "    return _f(FakeBackend(fixtures.fixture_8x_mi355x()))\n"
+    )
+    dev = create_device_from_plugin(str(plugin_file))
+    dev.start()
+    assert dev.get_name() == "amdgpu"
+
+
+def test_p2p_false_caps_bandwidth():
+    """An xGMI link with p2p disabled degrades to the host path."""
+    from kubegpu_amd.discovery import fixtures
+
+    fix = fixtures.fixture_8x_mi355x()
+    for l in fix.devices[0].links:
+        if l.peer_index == 1:
+            l.p2p = False
+    bw = fix.bandwidth_matrix()
+    assert bw[0][1] == 63.0
+    assert bw[0][2] == 153.0
