@@ -82,14 +82,31 @@ class Cluster:
         """Fit -> choose node -> allocate -> bind -> commit."""
         t0 = time.perf_counter()
         candidates: List[Tuple[Tuple[float, int], str, PodInfo, List[str]]] = []
+        # The topology-aware translation synthesizes against the
+        # cluster-wide canonical-tree cache, so it is node-independent
+        # (the flat knob==0 path wraps against per-node advertisement
+        # instead): translate once, bind per candidate node.
+        from ..scheduler.translate import GPU_TOPOLOGY_GENERATION
+
+        shared: Optional[PodInfo] = None
+        if pod.requests.get(GPU_TOPOLOGY_GENERATION) in (None, 1) and self.node_infos:
+            shared = pod.copy()
+            first_ni = self.node_infos[utils.sorted_string_keys(self.node_infos)[0]]
+            try:
+                self.scheduler.pod_allocate(first_ni, shared)
+            except SchedulingError:
+                raise SchedulingError(f"no node fits pod {pod.name}")
         for name in utils.sorted_string_keys(self.node_infos):
             ni = self.node_infos[name]
             # fit == "a translation + binding exists": the bind attempt
             # below subsumes the pod_fits_device predicate (which stays
             # available for API parity / external callers).
-            trial = pod.copy()
             try:
-                self.scheduler.pod_allocate(ni, trial)
+                if shared is not None:
+                    trial = shared.copy()
+                else:
+                    trial = pod.copy()
+                    self.scheduler.pod_allocate(ni, trial)
                 uuids = self.core.bind_pod(name, trial, commit=False)
             except SchedulingError:
                 continue
