@@ -111,6 +111,39 @@ def main() -> int:
 
     sched = run_sched_bench(args.pods) if rank == 0 else {}
 
+    # On a real GPU node, also schedule a k=n_gpus pod against the
+    # DISCOVERED topology and record the chosen set + the model's
+    # predicted ring bottleneck — the measured busbw below verifies it
+    # (SURVEY.md hard part (b): scoring must track RCCL throughput).
+    real_sched = {}
+    if rank == 0 and on_gpu:
+        try:
+            from kubegpu_amd.api.types import ContainerInfo, PodInfo
+            from kubegpu_amd.core import Cluster
+            from kubegpu_amd.deviceplugin import create_device_plugin
+            from kubegpu_amd.discovery import default_backend
+            from kubegpu_amd.plugintypes import RESOURCE_GPU
+
+            cluster = Cluster()
+            mgr = create_device_plugin(default_backend())
+            cluster.add_node_from_manager("local", mgr)
+            pod = PodInfo(
+                name=f"bench-{n_gpus}",
+                running_containers={
+                    "c": ContainerInfo(kube_requests={RESOURCE_GPU: n_gpus})
+                },
+            )
+            res = cluster.schedule(pod)
+            st = cluster.core.nodes["local"]
+            idxs = sorted(st.gpus[u].index for u in res.uuids)
+            pred = st.scorer.ring_bw(idxs)
+            real_sched = {
+                "scheduled_devices": idxs,
+                "predicted_ring_bottleneck_gbps": None if pred >= 1e9 else pred,
+            }
+        except Exception as e:  # never fail the bench on discovery hiccups
+            real_sched = {"scheduled_devices": None, "sched_error": str(e)[:200]}
+
     nbytes = args.bytes
     if world > 1:
         from kubegpu_amd.probe.rccl_probe import torch_allreduce_busbw
@@ -178,6 +211,7 @@ def main() -> int:
                 "mode": mode,
                 "buffer_bytes": nbytes,
                 "scenario": f"BASELINE.json k={n_gpus}",
+                **real_sched,
             },
             **sched,
         }
