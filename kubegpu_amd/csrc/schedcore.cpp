@@ -119,22 +119,34 @@ std::tuple<double, int, double> score_subset(const std::vector<int>& sub,
 
 // Candidates are positions 0..n-1 (the Python wrapper maps real GPU
 // indices); lexicographic combination order + strictly-greater keeps the
-// same winner as itertools.combinations in xgmi.py.
-std::vector<int> choose_best_subset(int n, int k, const std::vector<double>& bw_flat) {
+// same winner as itertools.combinations in xgmi.py.  `must` (positions)
+// constrains the search to supersets of that set — the kubelet
+// GetPreferredAllocation contract (must_include_deviceIDs).
+std::vector<int> choose_best_subset(int n, int k, const std::vector<double>& bw_flat,
+                                    const std::vector<int>& must) {
   const double* bw = bw_flat.data();
   std::vector<int> free(n);
   for (int i = 0; i < n; ++i) free[i] = i;
-  if (k <= 0 || k > n) return {};
+  if (k <= 0 || k > n || (int)must.size() > k) return {};
   if (k == n) return free;
+  uint64_t must_mask = 0;
+  for (int g : must) {
+    if (g < 0 || g >= n) return {};
+    must_mask |= (uint64_t)1 << g;
+  }
   std::vector<int> comb(k);
   for (int i = 0; i < k; ++i) comb[i] = i;
   std::tuple<double, int, double> best{-1.0, -1, -1.0};
   std::vector<int> best_sub;
   while (true) {
-    auto s = score_subset(comb, free, bw, n);
-    if (s > best) {
-      best = s;
-      best_sub = comb;
+    uint64_t mask = 0;
+    for (int g : comb) mask |= (uint64_t)1 << g;
+    if ((mask & must_mask) == must_mask) {
+      auto s = score_subset(comb, free, bw, n);
+      if (s > best) {
+        best = s;
+        best_sub = comb;
+      }
     }
     // next lexicographic combination
     int i = k - 1;
@@ -158,7 +170,7 @@ std::pair<double, std::vector<int>> best_ring_py(int n, const std::vector<int>& 
 PYBIND11_MODULE(_schedcore, m) {
   m.doc() = "native xGMI subset scorer (twin of kubegpu_amd.scheduler.xgmi)";
   m.def("choose_best_subset", &choose_best_subset, py::arg("n"), py::arg("k"),
-        py::arg("bw_flat"));
+        py::arg("bw_flat"), py::arg("must") = std::vector<int>());
   m.def("best_ring", &best_ring_py, py::arg("n"), py::arg("subset"),
         py::arg("bw_flat"));
 }

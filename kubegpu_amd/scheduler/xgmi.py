@@ -130,22 +130,29 @@ def score_subset(
 
 
 def choose_best_subset(
-    free: Sequence[int], k: int, bw: BwMatrix
+    free: Sequence[int], k: int, bw: BwMatrix, must: Sequence[int] = ()
 ) -> List[int]:
     """Best k-subset of *free*: max ring bandwidth, then least
     fragmentation of the remainder, then max aggregate ring bandwidth,
-    then lexicographically smallest (determinism).
+    then lexicographically smallest (determinism).  *must* constrains
+    the search to supersets of that set (kubelet's
+    must_include_deviceIDs contract).
 
-    Returns [] when k > len(free).
+    Returns [] when k > len(free) or the constraint is unsatisfiable.
     """
     free_sorted = sorted(free)
-    if k <= 0 or k > len(free_sorted):
+    must_set = set(must)
+    if k <= 0 or k > len(free_sorted) or len(must_set) > k:
+        return []
+    if not must_set.issubset(free_sorted):
         return []
     if k == len(free_sorted):
         return free_sorted
     best: Tuple[float, int, float] = (-1.0, -1, -1.0)
     best_sub: List[int] = []
     for sub in itertools.combinations(free_sorted, k):
+        if not must_set.issubset(sub):
+            continue
         s = score_subset(sub, free_sorted, bw)
         if s > best:
             best = s
@@ -185,17 +192,24 @@ class TopologyScorer:
         self._ring_memo: Dict[Tuple, float] = {}
         self._edges_memo: Dict[Tuple, int] = {}
 
-    def choose(self, free: Sequence[int], k: int) -> List[int]:
-        key = (tuple(sorted(free)), k)
+    def choose(
+        self, free: Sequence[int], k: int, must: Sequence[int] = ()
+    ) -> List[int]:
+        must_t = tuple(sorted(set(must)))
+        key = (tuple(sorted(free)), k, must_t)
         hit = self._choose_memo.get(key)
         if hit is not None:
             return list(hit)
         free_sorted = sorted(free)
-        if k <= 0 or k > len(free_sorted):
+        if k <= 0 or k > len(free_sorted) or len(must_t) > k:
             out: List[int] = []
+        elif not set(must_t).issubset(free_sorted):
+            out = []
         elif self._native is not None:
             if len(free_sorted) == self.n:
-                picked = self._native.choose_best_subset(self.n, k, self.flat)
+                picked = self._native.choose_best_subset(
+                    self.n, k, self.flat, [self.pos[g] for g in must_t]
+                )
                 out = [self.idx[p] for p in picked]
             else:
                 # sub-matrix for the current free set
@@ -206,10 +220,13 @@ class TopologyScorer:
                     for b in range(m):
                         if a != b:
                             sub[a * m + b] = self.flat[pa * self.n + self.pos[free_sorted[b]]]
-                picked = self._native.choose_best_subset(m, k, sub)
+                subpos = {g: i for i, g in enumerate(free_sorted)}
+                picked = self._native.choose_best_subset(
+                    m, k, sub, [subpos[g] for g in must_t]
+                )
                 out = [free_sorted[p] for p in picked]
         else:
-            out = choose_best_subset(free_sorted, k, self._bw)
+            out = choose_best_subset(free_sorted, k, self._bw, must_t)
         self._choose_memo[key] = out
         return list(out)
 

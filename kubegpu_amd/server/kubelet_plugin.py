@@ -102,7 +102,8 @@ class DevicePluginServicer:
             k = creq.allocation_size
             chosen: List[str]
             if scorer is None or k >= len(avail):
-                chosen = avail[:k]
+                # must_include devices lead, then the rest of avail
+                chosen = list(dict.fromkeys([*must, *avail]))[:k]
             else:
                 uuid_to_idx = {
                     u: self.manager.gpus[u].index
@@ -111,12 +112,14 @@ class DevicePluginServicer:
                 }
                 idx_to_uuid = {v: ku for ku, v in uuid_to_idx.items()}
                 free_idx = [uuid_to_idx[u] for u in avail if u in uuid_to_idx]
-                picked = scorer.choose(free_idx, k)
+                must_idx = [uuid_to_idx[m] for m in must if m in uuid_to_idx]
+                # best ring subset CONSTRAINED to contain the must-set
+                picked = scorer.choose(free_idx, k, must=must_idx)
+                if not picked and must_idx:
+                    # unsatisfiable constraint (e.g. must ⊄ avail): kubelet
+                    # contract still wants the musts honoured best-effort
+                    picked = scorer.choose(free_idx, k)
                 chosen = [idx_to_uuid[i] for i in picked]
-                # honour must_include (kubelet contract)
-                for m in must:
-                    if m not in chosen and chosen:
-                        chosen[-1] = m
             responses.append(dpapi.ContainerPreferredAllocationResponse(deviceIDs=chosen))
         return dpapi.PreferredAllocationResponse(container_responses=responses)
 

@@ -162,3 +162,45 @@ def test_registration_roundtrip(tmp_path):
     assert received["req"].endpoint == "amdgpu.sock"
     p.stop()
     server.stop(grace=0.5)
+
+
+def test_preferred_allocation_must_include(channel):
+    """must_include_deviceIDs are honoured: the chosen set contains every
+    must device and completes with that device's hive-mates."""
+    req = dpapi.PreferredAllocationRequest(
+        container_requests=[
+            dpapi.ContainerPreferredAllocationRequest(
+                available_deviceIDs=[f"GPU-mi355x-{i:02d}" for i in range(8)],
+                must_include_deviceIDs=["GPU-mi355x-05"],
+                allocation_size=2,
+            )
+        ]
+    )
+    resp = _stub(channel, "GetPreferredAllocation",
+                 dpapi.PreferredAllocationRequest,
+                 dpapi.PreferredAllocationResponse)(req, timeout=10)
+    ids = resp.container_responses[0].deviceIDs
+    idx = sorted(int(u.split("-")[-1]) for u in ids)
+    assert 5 in idx and len(idx) == 2
+    # completion stays in hive 1 (4-7): cross-hive would be PCIe-bound
+    assert all(i in (4, 5, 6, 7) for i in idx)
+
+
+def test_preferred_allocation_two_musts_both_kept(channel):
+    """Two must devices never collapse into one slot (the old last-slot
+    overwrite bug)."""
+    req = dpapi.PreferredAllocationRequest(
+        container_requests=[
+            dpapi.ContainerPreferredAllocationRequest(
+                available_deviceIDs=[f"GPU-mi355x-{i:02d}" for i in range(8)],
+                must_include_deviceIDs=["GPU-mi355x-00", "GPU-mi355x-05"],
+                allocation_size=4,
+            )
+        ]
+    )
+    resp = _stub(channel, "GetPreferredAllocation",
+                 dpapi.PreferredAllocationRequest,
+                 dpapi.PreferredAllocationResponse)(req, timeout=10)
+    idx = sorted(int(u.split("-")[-1])
+                 for u in resp.container_responses[0].deviceIDs)
+    assert len(idx) == 4 and 0 in idx and 5 in idx

@@ -4,6 +4,7 @@ import math
 
 from kubegpu_amd.discovery import fixtures
 from kubegpu_amd.scheduler.xgmi import (
+    TopologyScorer,
     best_ring,
     choose_best_subset,
     choose_best_subset_fast,
@@ -120,3 +121,34 @@ def test_degraded_mesh_avoids_down_links():
     hops = {tuple(sorted((order[i], order[(i + 1) % len(order)])))
             for i in range(len(order))}
     assert not (hops & down)
+
+
+def test_choose_with_must_constraint(fixture_2hive):
+    """must_include forces the subset to contain the given GPUs and the
+    chooser completes the set with the best-connected peers."""
+    bw = _bw(fixture_2hive)
+    # must=GPU5 (hive 1): the pair must be inside hive 1, containing 5
+    picked = choose_best_subset(list(range(8)), 2, bw, must=[5])
+    assert 5 in picked and all(g in (4, 5, 6, 7) for g in picked)
+    # must spanning both hives: still honoured, ring is PCIe-bound but
+    # the musts are in
+    picked = choose_best_subset(list(range(8)), 4, bw, must=[0, 5])
+    assert {0, 5}.issubset(picked)
+    # unsatisfiable: must larger than k, or must not in free
+    assert choose_best_subset(list(range(8)), 1, bw, must=[0, 5]) == []
+    assert choose_best_subset([1, 2, 3], 2, bw, must=[7]) == []
+
+
+def test_choose_with_must_native_matches_python(fixture_2hive):
+    """Native and Python must-constrained choosers agree (scorer path)."""
+    bw = _bw(fixture_2hive)
+    scorer = TopologyScorer(list(range(8)), bw)
+    for free, k, must in [
+        (list(range(8)), 2, [5]),
+        (list(range(8)), 4, [0, 5]),
+        ([2, 3, 4, 5, 6, 7], 3, [2]),
+        ([1, 3, 5, 7], 2, [3, 7]),
+    ]:
+        assert scorer.choose(free, k, must=must) == choose_best_subset(
+            free, k, bw, must=must
+        )
