@@ -96,7 +96,27 @@ class Cluster:
                 self.scheduler.pod_allocate(first_ni, shared)
             except SchedulingError:
                 raise SchedulingError(f"no node fits pod {pod.name}")
-        for name in utils.sorted_string_keys(self.node_infos):
+        # Equivalence-class dedup: nodes with identical topology
+        # fingerprint AND identical free-position sets produce identical
+        # bind results and scores, so only one representative per class
+        # needs a (relatively expensive) trial bind.  Representative =
+        # the class member the full sort below would have picked among
+        # its (tied) members: min name for the first-fit "naive" policy,
+        # max name for the reverse (score, name) sort of "xgmi".
+        ordered = utils.sorted_string_keys(self.node_infos)
+        reps: Dict[Tuple, str] = {}
+        for name in ordered:
+            sig = self.core.state_signature(name)
+            if sig is None:
+                sig = ("__unregistered__", name)
+            if self.policy == "naive":
+                reps.setdefault(sig, name)
+            else:
+                reps[sig] = name
+        rep_names = set(reps.values())
+        for name in ordered:
+            if name not in rep_names:
+                continue
             ni = self.node_infos[name]
             # fit == "a translation + binding exists": the bind attempt
             # below subsumes the pod_fits_device predicate (which stays

@@ -57,9 +57,36 @@ class NodeState:
     bw: BwMatrix = field(default_factory=dict)  # index -> index -> GB/s
     index_to_uuid: Dict[int, str] = field(default_factory=dict)
     scorer: Optional[TopologyScorer] = None
+    topo_token: int = -1  # interned structural fingerprint (shape + bw)
+    _free_sig: Optional[Tuple] = None  # cache; invalidated on used changes
 
     def free_uuids(self) -> List[str]:
         return [u for u in sorted(self.gpus) if u not in self.used]
+
+    def mark_used(self, uuid: str) -> None:
+        self.used.add(uuid)
+        self._free_sig = None
+
+    def mark_free(self, uuid: str) -> None:
+        self.used.discard(uuid)
+        self._free_sig = None
+
+    def free_position_sig(self) -> Tuple:
+        """Signature of the free set by (tree position, scoring index).
+
+        Two nodes with equal topo_token and equal free_position_sig are
+        score-equivalent for any pod: binding and subset scoring depend
+        only on group structure, the bandwidth matrix over indices, and
+        which positions are free — never on uuids or node names."""
+        if self._free_sig is None:
+            self._free_sig = tuple(
+                sorted(
+                    (g.h_pos, g.g_pos, g.index)
+                    for u, g in self.gpus.items()
+                    if u not in self.used
+                )
+            )
+        return self._free_sig
 
 
 def _synthetic_bw(gpus: List[_Gpu]) -> BwMatrix:
@@ -95,6 +122,10 @@ class GroupScheduler:
     def __init__(self, policy: str = "xgmi") -> None:
         self._lock = threading.RLock()
         self.nodes: Dict[str, NodeState] = {}
+        # topology-fingerprint interning: identical (shape, bw matrix)
+        # fingerprints share one small token so per-pod signature checks
+        # never re-hash the full matrix (state_signature hot path)
+        self._sig_intern: Dict[Tuple, int] = {}
         assert policy in ("xgmi", "naive")
         self.policy = policy
 
@@ -137,7 +168,22 @@ class GroupScheduler:
         else:
             state.bw = _synthetic_bw(list(state.gpus.values()))
         state.scorer = TopologyScorer(list(state.index_to_uuid.keys()), state.bw)
+        shape = tuple(
+            tuple(tuple(sorted(state.gpus[u].index for u in ids)) for _, ids in g_items)
+            for _, g_items in layout.groups
+        )
+        bw_fp = tuple(
+            sorted(
+                (i, j, round(v, 3))
+                for i, row in state.bw.items()
+                for j, v in row.items()
+            )
+        )
+        topo_sig = (shape, bw_fp)
         with self._lock:
+            state.topo_token = self._sig_intern.setdefault(
+                topo_sig, len(self._sig_intern)
+            )
             # node re-registration (watch update / re-discovery) must not
             # forget live allocations: carry over used flags for GPUs
             # that still exist (cf. the manager's in_use-survives-
@@ -254,7 +300,7 @@ class GroupScheduler:
                 cont.allocate_from[req] = concrete
             if commit:
                 for uuid in chosen_all:
-                    state.used.add(uuid)
+                    state.mark_used(uuid)
             return chosen_all
 
     def _match_positions(
@@ -362,7 +408,7 @@ class GroupScheduler:
             if state is None:
                 return
             for uuid in self._pod_uuids(pod):
-                state.used.add(uuid)
+                state.mark_used(uuid)
 
     def return_pod_resources(self, node_name: str, pod: PodInfo) -> None:
         with self._lock:
@@ -370,7 +416,7 @@ class GroupScheduler:
             if state is None:
                 return
             for uuid in self._pod_uuids(pod):
-                state.used.discard(uuid)
+                state.mark_free(uuid)
 
     @staticmethod
     def _pod_uuids(pod: PodInfo) -> Set[str]:
@@ -383,6 +429,17 @@ class GroupScheduler:
                 except ValueError:
                     pass
         return uuids
+
+    def state_signature(self, node_name: str) -> Optional[Tuple]:
+        """(topo_sig, free_position_sig) — equal signatures mean the two
+        nodes produce identical bind results and scores for any pod, so
+        a cluster scheduler need only try one representative per class
+        (O(distinct states) instead of O(nodes) bind attempts)."""
+        with self._lock:
+            state = self.nodes.get(node_name)
+            if state is None:
+                return None
+            return (state.topo_token, state.free_position_sig())
 
     def free_count(self, node_name: str) -> int:
         with self._lock:
