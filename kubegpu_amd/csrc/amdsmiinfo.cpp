@@ -52,6 +52,8 @@ struct Gpu {
   uint64_t vram_total_bytes = 0;
   std::string vram_type = "HBM3E";
   double vram_bandwidth_gbps = 0.0;
+  uint64_t ecc_correctable = 0;
+  uint64_t ecc_uncorrectable = 0;
   std::vector<Link> links;
   amdsmi_processor_handle handle{};
 };
@@ -167,6 +169,14 @@ bool collect(std::vector<Gpu>& gpus, std::string& driver_version) {
       if (amdsmi_get_gpu_topo_numa_affinity(h, &numa) == AMDSMI_STATUS_SUCCESS && numa >= 0)
         g.numa_node = numa;
 
+      // Accumulated RAS/ECC error totals: the node agent flips a GPU
+      // Unhealthy (kubelet ListAndWatch) on any uncorrectable error.
+      amdsmi_error_count_t ec{};
+      if (amdsmi_get_gpu_total_ecc_count(h, &ec) == AMDSMI_STATUS_SUCCESS) {
+        g.ecc_correctable = ec.correctable_count;
+        g.ecc_uncorrectable = ec.uncorrectable_count;
+      }
+
       if (driver_version.empty()) {
         amdsmi_driver_info_t dinfo{};
         if (amdsmi_get_gpu_driver_info(h, &dinfo) == AMDSMI_STATUS_SUCCESS)
@@ -230,6 +240,8 @@ void print_json(const std::vector<Gpu>& gpus, const std::string& driver) {
     printf("   \"card_path\": \"%s\",\n", json_escape(g.card_path).c_str());
     printf("   \"numa_node\": %d,\n", g.numa_node);
     printf("   \"compute_units\": %u,\n", g.compute_units);
+    printf("   \"ecc_correctable\": %" PRIu64 ",\n", g.ecc_correctable);
+    printf("   \"ecc_uncorrectable\": %" PRIu64 ",\n", g.ecc_uncorrectable);
     printf("   \"memory\": {\"vram_total_bytes\": %" PRIu64
            ", \"vram_type\": \"%s\", \"vram_bandwidth_gbps\": %.1f},\n",
            g.vram_total_bytes, json_escape(g.vram_type).c_str(), g.vram_bandwidth_gbps);
@@ -256,6 +268,8 @@ void print_human(const std::vector<Gpu>& gpus, const std::string& driver) {
     printf("  VRAM:   %.1f GiB %s (%.0f GB/s)\n", g.vram_total_bytes / 1073741824.0,
            g.vram_type.c_str(), g.vram_bandwidth_gbps);
     printf("  NUMA:   %d   CUs: %u\n", g.numa_node, g.compute_units);
+    printf("  ECC:    %" PRIu64 " correctable / %" PRIu64 " uncorrectable\n",
+           g.ecc_correctable, g.ecc_uncorrectable);
   }
   printf("\nPairwise topology (type/hops/weight/GBps/p2p):\n");
   for (const Gpu& g : gpus) {

@@ -49,6 +49,7 @@ from ..plugintypes import RESOURCE_GPU
 ALLOCATE_RE = re.compile(r".*/gpugrp1/[^/]+/gpugrp0/[^/]+/gpu/([^/]+)/cards$")
 
 DISCOVERY_CACHE_S = 300.0  # 5 minutes, like the reference
+VANISHED_TTL_S = 600.0  # how long a vanished GPU stays visible-Unhealthy
 
 
 class _UnionFind:
@@ -79,6 +80,11 @@ class AMDGPUManager(Device):
         self.index_to_id: Dict[int, str] = {}
         self._last_get_time: float = 0.0
         self._last_info: Optional[GpusInfo] = None
+        # Recently-vanished GPUs kept as tombstones so the kubelet
+        # device plugin can report them Unhealthy (capacity visible,
+        # allocatable 0) instead of silently shrinking the node, until
+        # the grace window expires: uuid -> (last GpuInfo, swept-at).
+        self.vanished: Dict[str, tuple] = {}
 
     # -- Device interface --------------------------------------------------
 
@@ -128,16 +134,43 @@ class AMDGPUManager(Device):
                 dev.found = True
                 dev.in_use = in_use
                 self.gpus[dev.uuid] = dev
-            # ...sweep.
+            # ...sweep (tombstone first, drop after the grace window).
             for uuid in [u for u, g in self.gpus.items() if not g.found]:
                 utils.logf(2, "GPU %s vanished; removing from inventory", uuid)
+                self.vanished[uuid] = (self.gpus[uuid], now)
                 del self.gpus[uuid]
+            for uuid in list(self.vanished):
+                if uuid in self.gpus:  # came back
+                    del self.vanished[uuid]
+                elif now - self.vanished[uuid][1] > VANISHED_TTL_S:
+                    del self.vanished[uuid]
 
             self.path_to_id = {g.render_path: u for u, g in self.gpus.items() if g.render_path}
             self.bdf_to_id = {g.bdf: u for u, g in self.gpus.items() if g.bdf}
             self.index_to_id = {g.index: u for u, g in self.gpus.items()}
 
             self._topology_discovery(info)
+
+    def device_health(self) -> Dict[str, bool]:
+        """uuid -> healthy for every advertisable device.
+
+        Present GPUs are healthy unless they report uncorrectable ECC
+        errors (GpuInfo.healthy); tombstoned (recently-vanished) GPUs
+        stay visible but always unhealthy so kubelet degrades
+        allocatable instead of the node silently shrinking."""
+        with self._lock:
+            out = {u: g.healthy for u, g in self.gpus.items()}
+            for u in self.vanished:
+                out.setdefault(u, False)
+            return out
+
+    def gpu_or_tombstone(self, uuid: str) -> Optional[GpuInfo]:
+        with self._lock:
+            g = self.gpus.get(uuid)
+            if g is not None:
+                return g
+            t = self.vanished.get(uuid)
+            return t[0] if t else None
 
     def _topology_discovery(self, info: GpusInfo) -> None:
         """Derive the two-level gpugrp names from the xGMI graph.

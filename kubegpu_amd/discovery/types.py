@@ -95,6 +95,8 @@ class GpuInfo:
     card_path: str = ""
     numa_node: int = 0
     compute_units: int = 0
+    ecc_correctable: int = 0
+    ecc_uncorrectable: int = 0
     memory: MemoryInfo = field(default_factory=MemoryInfo)
     links: List[LinkInfo] = field(default_factory=list)
 
@@ -104,6 +106,12 @@ class GpuInfo:
     in_use: bool = False
     topo_done: bool = False
     name: str = ""  # topology-prefixed: gpugrp1/H/gpugrp0/G/gpu/<uuid>
+
+    @property
+    def healthy(self) -> bool:
+        """Any uncorrectable (fatal) ECC error marks the GPU unhealthy;
+        correctable errors are informational only."""
+        return self.ecc_uncorrectable == 0
 
     def to_dict(self) -> dict:
         return {
@@ -117,6 +125,8 @@ class GpuInfo:
             "card_path": self.card_path,
             "numa_node": self.numa_node,
             "compute_units": self.compute_units,
+            "ecc_correctable": self.ecc_correctable,
+            "ecc_uncorrectable": self.ecc_uncorrectable,
             "memory": self.memory.to_dict(),
             "links": [l.to_dict() for l in self.links],
         }
@@ -134,6 +144,8 @@ class GpuInfo:
             card_path=str(d.get("card_path", "")),
             numa_node=int(d.get("numa_node", 0)),
             compute_units=int(d.get("compute_units", 0)),
+            ecc_correctable=int(d.get("ecc_correctable", 0)),
+            ecc_uncorrectable=int(d.get("ecc_uncorrectable", 0)),
             memory=MemoryInfo.from_dict(d.get("memory", {})),
             links=[LinkInfo.from_dict(x) for x in d.get("links", [])],
         )

@@ -55,16 +55,24 @@ class DevicePluginServicer:
         devices = []
         try:
             self.manager.update_gpu_info()
-            healthy = True
+            discovery_ok = True
         except DiscoveryError:
-            healthy = False
-        for uuid in sorted(self.manager.gpus):
-            g = self.manager.gpus[uuid]
-            topo = dpapi.TopologyInfo(nodes=[dpapi.NUMANode(ID=g.numa_node)])
+            discovery_ok = False
+        # Per-device health: present GPUs with uncorrectable ECC errors
+        # and recently-vanished (tombstoned) GPUs are advertised
+        # UNHEALTHY — kubelet keeps the capacity visible but stops
+        # allocating — instead of the node silently shrinking.
+        health = self.manager.device_health()
+        for uuid in sorted(health):
+            g = self.manager.gpu_or_tombstone(uuid)
+            numa = g.numa_node if g is not None else 0
+            topo = dpapi.TopologyInfo(nodes=[dpapi.NUMANode(ID=numa)])
             devices.append(
                 dpapi.Device(
                     ID=uuid,
-                    health=dpapi.HEALTHY if healthy else dpapi.UNHEALTHY,
+                    health=dpapi.HEALTHY
+                    if discovery_ok and health[uuid]
+                    else dpapi.UNHEALTHY,
                     topology=topo,
                 )
             )

@@ -204,3 +204,69 @@ def test_preferred_allocation_two_musts_both_kept(channel):
     idx = sorted(int(u.split("-")[-1])
                  for u in resp.container_responses[0].deviceIDs)
     assert len(idx) == 4 and 0 in idx and 5 in idx
+
+
+def test_ecc_unhealthy_device_flagged(tmp_path):
+    """A GPU with uncorrectable ECC errors is advertised UNHEALTHY;
+    healthy peers stay HEALTHY."""
+    from kubegpu_amd.discovery import FakeBackend, fixtures
+
+    fix = fixtures.fixture_8x_mi355x()
+    fix.devices[3].ecc_uncorrectable = 2
+    mgr = create_device_plugin(FakeBackend(fix))
+    mgr.start()
+    p = KubeletDevicePlugin(mgr, socket_path=str(tmp_path / "ecc.sock"))
+    p.start()
+    try:
+        ch = grpc.insecure_channel(f"unix://{p.socket_path}")
+        stream = _stub(ch, "ListAndWatch", dpapi.Empty,
+                       dpapi.ListAndWatchResponse, streaming=True)(
+            dpapi.Empty(), timeout=10
+        )
+        frame = next(iter(stream))
+        by_id = {d.ID: d.health for d in frame.devices}
+        assert len(by_id) == 8
+        assert by_id["GPU-mi355x-03"] == dpapi.UNHEALTHY
+        assert by_id["GPU-mi355x-00"] == dpapi.HEALTHY
+        ch.close()
+    finally:
+        p.stop()
+
+
+def test_vanished_gpu_tombstoned_unhealthy(tmp_path):
+    """A GPU that disappears between discoveries stays advertised as an
+    UNHEALTHY device (tombstone) instead of silently shrinking the node."""
+    from kubegpu_amd.discovery import FakeBackend, fixtures
+
+    full = fixtures.fixture_8x_mi355x()
+    backend = FakeBackend(full)
+    mgr = create_device_plugin(backend)
+    mgr.start()
+    assert len(mgr.gpus) == 8
+    shrunk = fixtures.fixture_8x_mi355x()
+    gone = shrunk.devices.pop(5)
+    backend.set_info(shrunk)
+    mgr.update_gpu_info(force=True)
+    assert len(mgr.gpus) == 7 and gone.uuid in mgr.vanished
+    health = mgr.device_health()
+    assert health[gone.uuid] is False
+    assert sum(health.values()) == 7
+
+    p = KubeletDevicePlugin(mgr, socket_path=str(tmp_path / "van.sock"))
+    p.start()
+    try:
+        ch = grpc.insecure_channel(f"unix://{p.socket_path}")
+        frame = next(iter(_stub(ch, "ListAndWatch", dpapi.Empty,
+                                dpapi.ListAndWatchResponse, streaming=True)(
+            dpapi.Empty(), timeout=10)))
+        by_id = {d.ID: d.health for d in frame.devices}
+        assert len(by_id) == 8  # tombstone still advertised
+        assert by_id[gone.uuid] == dpapi.UNHEALTHY
+        # ...and it comes back HEALTHY when re-discovered
+        backend.set_info(fixtures.fixture_8x_mi355x())
+        mgr.update_gpu_info(force=True)
+        assert gone.uuid not in mgr.vanished
+        assert mgr.device_health()[gone.uuid] is True
+        ch.close()
+    finally:
+        p.stop()
