@@ -235,19 +235,27 @@ class AMDGPUManager(Device):
                 rl[RESOURCE_GPU] = 0
             return
         with self._lock:
+            # capacity counts every discovered GPU; allocatable excludes
+            # ECC-unhealthy ones (a GPU with uncorrectable errors stays
+            # visible but must not take new pods — mirrors the kubelet
+            # plugin's per-device Unhealthy advertisement)
             count = len(self.gpus)
-            for rl in (node_info.capacity, node_info.allocatable,
-                       node_info.kube_cap, node_info.kube_alloc):
+            healthy_count = sum(1 for g in self.gpus.values() if g.healthy)
+            for rl in (node_info.capacity, node_info.kube_cap):
                 rl[RESOURCE_GPU] = count
+            for rl in (node_info.allocatable, node_info.kube_alloc):
+                rl[RESOURCE_GPU] = healthy_count
             for uuid in utils.sorted_string_keys(self.gpus):
                 gpu = self.gpus[uuid]
                 if not gpu.name:
                     continue
                 add_group_resource(node_info.capacity, f"{gpu.name}/cards", 1)
-                add_group_resource(node_info.allocatable, f"{gpu.name}/cards", 1)
                 add_group_resource(
                     node_info.capacity, f"{gpu.name}/memory", gpu.memory.vram_total_bytes
                 )
+                if not gpu.healthy:
+                    continue
+                add_group_resource(node_info.allocatable, f"{gpu.name}/cards", 1)
                 add_group_resource(
                     node_info.allocatable, f"{gpu.name}/memory", gpu.memory.vram_total_bytes
                 )

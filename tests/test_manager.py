@@ -175,3 +175,37 @@ def test_p2p_false_caps_bandwidth():
     bw = fix.bandwidth_matrix()
     assert bw[0][1] == 63.0
     assert bw[0][2] == 153.0
+
+
+def test_ecc_unhealthy_excluded_from_allocatable():
+    """Uncorrectable-ECC GPU: stays in capacity, leaves allocatable, and
+    the scheduler never places a pod on it."""
+    from kubegpu_amd.api.types import ContainerInfo, NodeInfo, PodInfo
+    from kubegpu_amd.core import Cluster
+    from kubegpu_amd.plugintypes import RESOURCE_GPU
+
+    fix = fixtures.fixture_8x_mi355x()
+    fix.devices[2].ecc_uncorrectable = 1
+    mgr = create_device_plugin(FakeBackend(fix))
+    mgr.start()
+    ni = NodeInfo(name="n")
+    mgr.update_node_info(ni)
+    assert ni.capacity[RESOURCE_GPU] == 8
+    assert ni.allocatable[RESOURCE_GPU] == 7
+    bad = fix.devices[2].uuid
+    assert any(bad in k for k in ni.capacity)
+    assert not any(bad in k for k in ni.allocatable)
+
+    cluster = Cluster()
+    cluster.add_node(ni, mgr._last_info, mgr)
+    seen = set()
+    pods = []
+    for i in range(3):
+        pod = PodInfo(
+            name=f"p{i}",
+            running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: 2})},
+        )
+        res = cluster.schedule(pod)
+        pods.append(pod)
+        seen.update(res.uuids)
+    assert bad not in seen and len(seen) == 6
