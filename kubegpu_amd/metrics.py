@@ -31,6 +31,7 @@ class Metrics:
         self.schedule_latencies: List[float] = []
         self.allocations = 0
         self.failures = 0
+        self.gpu_health: Dict[str, bool] = {}
         if _HAVE_PROM:
             # per-instance registry: multiple Metrics objects (tests,
             # embedded schedulers) must not collide in the global one
@@ -51,6 +52,18 @@ class Metrics:
             )
             self._xgmi = Gauge(
                 "kubegpu_amd_xgmi_link_gbps", "last probed xGMI ring bandwidth GB/s",
+                registry=self.registry,
+            )
+            self._gpu_health = Gauge(
+                "kubegpu_amd_gpu_healthy",
+                "1 = GPU healthy, 0 = unhealthy (ECC/vanished)",
+                ["uuid"],
+                registry=self.registry,
+            )
+            self._gpu_ecc = Gauge(
+                "kubegpu_amd_gpu_ecc_uncorrectable",
+                "accumulated uncorrectable ECC errors",
+                ["uuid"],
                 registry=self.registry,
             )
 
@@ -75,6 +88,13 @@ class Metrics:
     def set_xgmi_gbps(self, gbps: float) -> None:
         if _HAVE_PROM:
             self._xgmi.set(gbps)
+
+    def set_gpu_health(self, uuid: str, healthy: bool, ecc_uncorrectable: int = 0) -> None:
+        with self._lock:
+            self.gpu_health[uuid] = healthy
+        if _HAVE_PROM:
+            self._gpu_health.labels(uuid=uuid).set(1.0 if healthy else 0.0)
+            self._gpu_ecc.labels(uuid=uuid).set(float(ecc_uncorrectable))
 
     def percentile(self, q: float) -> Optional[float]:
         with self._lock:

@@ -89,6 +89,11 @@ def main(argv=None) -> int:
             plugin.start()
             registered = False
         plugin.servicer.notify()  # wake ListAndWatch to refresh health
+        for uuid, ok in manager.device_health().items():
+            g = manager.gpu_or_tombstone(uuid)
+            METRICS.set_gpu_health(
+                uuid, ok, g.ecc_uncorrectable if g is not None else 0
+            )
         time.sleep(args.health_interval if registered else 5.0)
 
     plugin.stop()
