@@ -42,6 +42,11 @@ def test_amdsmiinfo_binary_json():
     assert g.memory.vram_total_bytes > 200 * 1024**3
     assert g.render_path.startswith("/dev/dri/renderD")
     assert os.path.exists(g.render_path)
+    # ECC totals present in the payload (health source); a freshly
+    # provisioned box should have no uncorrectable errors
+    raw = json.loads(out.stdout.decode())
+    assert "ecc_uncorrectable" in raw["devices"][0]
+    assert g.ecc_uncorrectable == 0 and g.healthy
 
 
 def test_amdsmiinfo_human_mode():
