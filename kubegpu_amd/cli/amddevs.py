@@ -8,6 +8,7 @@ Modes:
   amddevs --schedule K    schedule a synthetic K-GPU pod against the local
                           node and print the chosen GPU set + allocation
   amddevs --probe K       same, then run the in-pod RCCL probe over the set
+  amddevs --health        per-GPU health view (ECC totals, tombstones)
 """
 
 from __future__ import annotations
@@ -29,9 +30,33 @@ def main(argv=None) -> int:
     p.add_argument("--schedule", type=int, metavar="K", help="schedule a K-GPU pod")
     p.add_argument("--probe", type=int, metavar="K", help="schedule + RCCL probe")
     p.add_argument("--bytes", type=int, default=256 << 20)
+    p.add_argument("--health", action="store_true",
+                   help="per-GPU health view (ECC totals, tombstones)")
+    p.add_argument("--fake", action="store_true",
+                   help="use the 8xMI355X fixture backend (no GPU needed)")
     args = p.parse_args(argv)
 
-    backend = default_backend()
+    if args.fake:
+        from ..discovery import FakeBackend, fixtures
+
+        backend = FakeBackend(fixtures.fixture_8x_mi355x())
+    else:
+        backend = default_backend()
+    if args.health:
+        mgr = create_device_plugin(backend)
+        mgr.start()
+        health = mgr.device_health()
+        rows = {}
+        for uuid in sorted(health):
+            g = mgr.gpu_or_tombstone(uuid)
+            rows[uuid] = {
+                "healthy": health[uuid],
+                "present": uuid in mgr.gpus,
+                "ecc_correctable": g.ecc_correctable if g else None,
+                "ecc_uncorrectable": g.ecc_uncorrectable if g else None,
+            }
+        print(json.dumps(rows, indent=1))
+        return 0
     if not (args.plugin or args.schedule or args.probe):
         print(backend.get_gpu_info().decode())
         return 0

@@ -296,3 +296,19 @@ def test_schedule_latency_flat_at_scale():
             cluster.release(live.pop(0))
     lat.sort()
     assert lat[int(0.95 * len(lat))] < 0.050  # generous CI bound; ~1ms measured
+
+
+def test_amddevs_cli_fake_modes(capsys):
+    """CLI smoke on CPU via --fake: health view and schedule mode."""
+    import json as _json
+
+    from kubegpu_amd.cli.amddevs import main as cli_main
+
+    assert cli_main(["--fake", "--health"]) == 0
+    rows = _json.loads(capsys.readouterr().out)
+    assert len(rows) == 8 and all(r["healthy"] for r in rows.values())
+
+    assert cli_main(["--fake", "--schedule", "4"]) == 0
+    out = _json.loads(capsys.readouterr().out)
+    assert len(out["gpus"]) == 4 and "/dev/kfd" in out["devices"]
+    assert out["envs"]["ROCR_VISIBLE_DEVICES"].count(",") == 3
