@@ -54,6 +54,7 @@ def main(argv=None) -> int:
                 "present": uuid in mgr.gpus,
                 "ecc_correctable": g.ecc_correctable if g else None,
                 "ecc_uncorrectable": g.ecc_uncorrectable if g else None,
+                "process_count": g.process_count if g else None,
             }
         print(json.dumps(rows, indent=1))
         return 0

@@ -54,6 +54,7 @@ struct Gpu {
   double vram_bandwidth_gbps = 0.0;
   uint64_t ecc_correctable = 0;
   uint64_t ecc_uncorrectable = 0;
+  uint32_t process_count = 0;
   std::vector<Link> links;
   amdsmi_processor_handle handle{};
 };
@@ -177,6 +178,13 @@ bool collect(std::vector<Gpu>& gpus, std::string& driver_version) {
         g.ecc_uncorrectable = ec.uncorrectable_count;
       }
 
+      // Compute processes currently on the GPU (the reference tracked
+      // InUse only from its own allocations; this sees external users).
+      uint32_t nproc = 0;
+      amdsmi_status_t pst = amdsmi_get_gpu_process_list(h, &nproc, nullptr);
+      if (pst == AMDSMI_STATUS_SUCCESS || pst == AMDSMI_STATUS_OUT_OF_RESOURCES)
+        g.process_count = nproc;
+
       if (driver_version.empty()) {
         amdsmi_driver_info_t dinfo{};
         if (amdsmi_get_gpu_driver_info(h, &dinfo) == AMDSMI_STATUS_SUCCESS)
@@ -242,6 +250,7 @@ void print_json(const std::vector<Gpu>& gpus, const std::string& driver) {
     printf("   \"compute_units\": %u,\n", g.compute_units);
     printf("   \"ecc_correctable\": %" PRIu64 ",\n", g.ecc_correctable);
     printf("   \"ecc_uncorrectable\": %" PRIu64 ",\n", g.ecc_uncorrectable);
+    printf("   \"process_count\": %u,\n", g.process_count);
     printf("   \"memory\": {\"vram_total_bytes\": %" PRIu64
            ", \"vram_type\": \"%s\", \"vram_bandwidth_gbps\": %.1f},\n",
            g.vram_total_bytes, json_escape(g.vram_type).c_str(), g.vram_bandwidth_gbps);
@@ -270,6 +279,7 @@ void print_human(const std::vector<Gpu>& gpus, const std::string& driver) {
     printf("  NUMA:   %d   CUs: %u\n", g.numa_node, g.compute_units);
     printf("  ECC:    %" PRIu64 " correctable / %" PRIu64 " uncorrectable\n",
            g.ecc_correctable, g.ecc_uncorrectable);
+    printf("  procs:  %u\n", g.process_count);
   }
   printf("\nPairwise topology (type/hops/weight/GBps/p2p):\n");
   for (const Gpu& g : gpus) {

@@ -97,6 +97,7 @@ class GpuInfo:
     compute_units: int = 0
     ecc_correctable: int = 0
     ecc_uncorrectable: int = 0
+    process_count: int = 0  # compute processes seen by amdsmi (external in-use)
     memory: MemoryInfo = field(default_factory=MemoryInfo)
     links: List[LinkInfo] = field(default_factory=list)
 
@@ -127,6 +128,7 @@ class GpuInfo:
             "compute_units": self.compute_units,
             "ecc_correctable": self.ecc_correctable,
             "ecc_uncorrectable": self.ecc_uncorrectable,
+            "process_count": self.process_count,
             "memory": self.memory.to_dict(),
             "links": [l.to_dict() for l in self.links],
         }
@@ -146,6 +148,7 @@ class GpuInfo:
             compute_units=int(d.get("compute_units", 0)),
             ecc_correctable=int(d.get("ecc_correctable", 0)),
             ecc_uncorrectable=int(d.get("ecc_uncorrectable", 0)),
+            process_count=int(d.get("process_count", 0)),
             memory=MemoryInfo.from_dict(d.get("memory", {})),
             links=[LinkInfo.from_dict(x) for x in d.get("links", [])],
         )
