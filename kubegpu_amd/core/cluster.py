@@ -41,6 +41,13 @@ class Cluster:
 
     def __init__(self, scheduler: Optional[AMDGPUScheduler] = None,
                  policy: str = "xgmi"):
+        import threading
+
+        # One pod binds at a time (the reference's core serializes
+        # scheduler-plugin calls; without this the trial
+        # bind and the commit race between threads and two pods can
+        # claim the same GPUs).
+        self._sched_lock = threading.Lock()
         self.scheduler = scheduler or AMDGPUScheduler()
         self.core = GroupScheduler(policy=policy)
         self.policy = policy
@@ -80,6 +87,10 @@ class Cluster:
 
     def schedule(self, pod: PodInfo) -> ScheduleResult:
         """Fit -> choose node -> allocate -> bind -> commit."""
+        with self._sched_lock:
+            return self._schedule_locked(pod)
+
+    def _schedule_locked(self, pod: PodInfo) -> ScheduleResult:
         t0 = time.perf_counter()
         candidates: List[Tuple[Tuple[float, int], str, PodInfo, List[str]]] = []
         # The topology-aware translation synthesizes against the

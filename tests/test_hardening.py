@@ -167,3 +167,49 @@ def test_zero_gpu_pod_schedules_nowhere_needed():
     pod = PodInfo(name="nogpu", running_containers={"c": ContainerInfo()})
     res = cluster.schedule(pod)
     assert res.uuids == []
+
+
+def test_no_double_allocation_under_contention():
+    """Threads racing for a scarce node must never hold overlapping GPU
+    sets simultaneously (TOCTOU between trial bind and commit)."""
+    cluster = Cluster()
+    mgr = create_device_plugin(FakeBackend(fixtures.fixture_8x_mi355x()))
+    cluster.add_node_from_manager("n0", mgr)
+    errors = []
+    overlap = []
+    held = {}
+    hlock = threading.Lock()
+
+    def worker(tid):
+        try:
+            for i in range(40):
+                pod = PodInfo(
+                    name=f"t{tid}-{i}",
+                    running_containers={
+                        "c": ContainerInfo(kube_requests={RESOURCE_GPU: 4})
+                    },
+                )
+                try:
+                    res = cluster.schedule(pod)
+                except SchedulingError:
+                    continue
+                mine = set(res.uuids)
+                with hlock:
+                    for other_tid, other in held.items():
+                        if other & mine:
+                            overlap.append((tid, other_tid, other & mine))
+                    held[tid] = mine
+                with hlock:
+                    held.pop(tid, None)
+                cluster.release(pod)
+        except Exception as e:  # pragma: no cover
+            errors.append(e)
+
+    threads = [threading.Thread(target=worker, args=(t,)) for t in range(6)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert not errors
+    assert not overlap, f"double-allocated GPUs: {overlap[:3]}"
+    assert cluster.core.free_count("n0") == 8
