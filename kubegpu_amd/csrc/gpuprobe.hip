@@ -82,6 +82,20 @@ __global__ void copy_kernel_v4_nt_chunk(const uint4v* __restrict__ src,
   }
 }
 
+// Mixed variant: regular (cache-allocating) loads + nontemporal stores.
+// Answers whether letting the read stream allocate in the 256 MiB LLC
+// helps or hurts a pure 1 GiB stream (no reuse => expected ~parity,
+// measured to settle it).
+__global__ void copy_kernel_v4_mixed(const uint4v* __restrict__ src,
+                                     uint4v* __restrict__ dst, size_t n4) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n4; i += stride) {
+    uint4v v = src[i];
+    __builtin_nontemporal_store(v, &dst[i]);
+  }
+}
+
 __global__ void copy_kernel_b(const unsigned char* __restrict__ src,
                               unsigned char* __restrict__ dst, size_t n) {
   size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -161,6 +175,10 @@ double copy_bw_gbps(int64_t nbytes, int64_t iters, int64_t blocks_arg,
                          (const uint4v*)src.data_ptr(), (uint4v*)dst.data_ptr(), n4);
     else if (variant == 2)
       hipLaunchKernelGGL(copy_kernel_v4_nt_chunk, dim3(blocks), dim3(BLOCK), 0,
+                         stream, (const uint4v*)src.data_ptr(),
+                         (uint4v*)dst.data_ptr(), n4);
+    else if (variant == 3)
+      hipLaunchKernelGGL(copy_kernel_v4_mixed, dim3(blocks), dim3(BLOCK), 0,
                          stream, (const uint4v*)src.data_ptr(),
                          (uint4v*)dst.data_ptr(), n4);
     else if (nontemporal)

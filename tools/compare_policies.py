@@ -44,6 +44,19 @@ def build_cluster(policy: str, n_nodes: int, topology: str, rng) -> Cluster:
     for i in range(n_nodes):
         if topology == "2hive":
             fix = fixtures.fixture_2hive_8gpu()
+        elif topology == "mixedfleet":
+            # heterogeneous fleet: healthy full-mesh, partitioned 2-hive
+            # and link-degraded nodes side by side — the regime where
+            # cross-node choice (which the reference lacks entirely:
+            # first-fit, score 0.0) dominates placement quality
+            kind = i % 3
+            if kind == 0:
+                fix = fixtures.fixture_8x_mi355x()
+            elif kind == 1:
+                fix = fixtures.fixture_2hive_8gpu()
+            else:
+                all_pairs = [(a, b) for a in range(8) for b in range(a + 1, 8)]
+                fix = fixtures.fixture_degraded_mesh(rng.sample(all_pairs, k=8))
         else:  # degraded: random subset of links down per node
             all_pairs = [(a, b) for a in range(8) for b in range(a + 1, 8)]
             missing = rng.sample(all_pairs, k=6)
@@ -101,7 +114,8 @@ def main() -> int:
     ap.add_argument("--pods", type=int, default=2000)
     ap.add_argument("--nodes", type=int, default=4)
     ap.add_argument("--seed", type=int, default=7)
-    ap.add_argument("--topology", choices=("2hive", "degraded"), default="degraded")
+    ap.add_argument("--topology", choices=("2hive", "degraded", "mixedfleet"),
+                    default="degraded")
     args = ap.parse_args()
 
     rng = random.Random(args.seed)
