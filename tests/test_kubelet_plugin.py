@@ -270,3 +270,47 @@ def test_vanished_gpu_tombstoned_unhealthy(tmp_path):
         ch.close()
     finally:
         p.stop()
+
+
+def test_register_with_fake_kubelet(tmp_path):
+    """register_with_kubelet posts a well-formed RegisterRequest to the
+    kubelet Registration service (verified against a fake kubelet)."""
+    import threading
+    from concurrent import futures
+
+    got = {}
+    done = threading.Event()
+
+    def register_handler(request, context):
+        got["version"] = request.version
+        got["endpoint"] = request.endpoint
+        got["resource_name"] = request.resource_name
+        got["preferred"] = request.options.get_preferred_allocation_available
+        done.set()
+        return dpapi.Empty()
+
+    kubelet_sock = str(tmp_path / "kubelet.sock")
+    server = grpc.server(futures.ThreadPoolExecutor(max_workers=2))
+    handler = grpc.method_handlers_generic_handler(
+        dpapi.REGISTRATION_SERVICE,
+        {"Register": grpc.unary_unary_rpc_method_handler(
+            register_handler,
+            request_deserializer=dpapi.RegisterRequest.FromString,
+            response_serializer=lambda m: m.SerializeToString(),
+        )},
+    )
+    server.add_generic_rpc_handlers((handler,))
+    server.add_insecure_port(f"unix://{kubelet_sock}")
+    server.start()
+    try:
+        mgr = create_device_plugin(FakeBackend(fixtures.fixture_8x_mi355x()))
+        mgr.start()
+        p = KubeletDevicePlugin(mgr, socket_path=str(tmp_path / "amdgpu.sock"))
+        p.register_with_kubelet(kubelet_sock)
+        assert done.wait(5)
+        assert got["version"] == dpapi.VERSION
+        assert got["endpoint"] == "amdgpu.sock"
+        assert got["resource_name"] == "amd.com/gpu"
+        assert got["preferred"] is True
+    finally:
+        server.stop(grace=0.5)
