@@ -16,6 +16,7 @@ contract is defined in this module and honoured by scheduler synthesis
 
 from __future__ import annotations
 
+import functools
 import re
 from typing import Tuple
 
@@ -72,10 +73,14 @@ def translate_resource(
 CARDS_RE = re.compile(r"^(?P<prefix>.*)/gpugrp1/(?P<h>[^/]+)/gpugrp0/(?P<g>[^/]+)/gpu/(?P<id>[^/]+)/cards$")
 
 
+@functools.lru_cache(maxsize=8192)
 def parse_cards_name(name: str) -> Tuple[str, str, str, str]:
     """Split an advertised/requested cards name into (prefix, h, g, id).
 
     Raises ValueError when the name is not a 2-level cards name.
+    Memoized: the schedule hot path re-parses the same synthesized and
+    advertised names constantly (names are drawn from a small set per
+    cluster, so an LRU holds the working set).
     """
     m = CARDS_RE.match(name)
     if not m:

@@ -126,6 +126,11 @@ class GroupScheduler:
         # fingerprints share one small token so per-pod signature checks
         # never re-hash the full matrix (state_signature hot path)
         self._sig_intern: Dict[Tuple, int] = {}
+        # bind-plan cache: binding is a pure function of (topology token,
+        # free-position set, demand shape) at the *index* level — uuids
+        # and names only enter when a plan is applied to a node.  Value:
+        # ordered [(demand key, [picked indices])], or None = infeasible.
+        self._plan_cache: Dict[Tuple, Optional[List[Tuple[Tuple, List[int]]]]] = {}
         assert policy in ("xgmi", "naive")
         self.policy = policy
 
@@ -244,37 +249,33 @@ class GroupScheduler:
             chosen_all: List[str] = []
             bindings: List[Tuple[ContainerInfo, str, str]] = []
 
-            # positions first (they are the topology-aware constraints)
-            pos_keys = [k for k in demands if k != (WILDCARD, WILDCARD)]
-            if pos_keys:
-                his: Dict[int, Dict[int, int]] = {}
-                for hi, gi in pos_keys:
-                    his.setdefault(hi, {})[gi] = len(demands[(hi, gi)])
-                assignment = self._match_positions(state, his, free, pod.name)
-                for (hi, gi), (ah, ag) in sorted(assignment.items()):
-                    members = self._group_members(state, ah, ag)
-                    cand = [state.gpus[u].index for u in members if u in free]
-                    k = len(demands[(hi, gi)])
-                    picked = self._choose(state, cand, k)
-                    picked_uuids = [state.index_to_uuid[i] for i in picked]
-                    for (cont, req), uuid in zip(demands[(hi, gi)], picked_uuids):
-                        bindings.append((cont, req, state.gpus[uuid].concrete_name))
-                        free.discard(uuid)
-                        chosen_all.append(uuid)
+            # Plan cache: the index-level choice is a pure function of
+            # (topology token, free positions, demand shape); None caches
+            # a deterministic infeasibility.  Applying a plan maps
+            # indices to this node's uuids/names.
+            plan_key = (
+                state.topo_token,
+                state.free_position_sig(),
+                tuple(sorted((k, len(v)) for k, v in demands.items())),
+            )
+            plan = self._plan_cache.get(plan_key, False)
+            if plan is None:
+                raise SchedulingError(
+                    f"node {node_name}: no group assignment satisfies pod {pod.name}"
+                )
+            if plan is False:
+                if len(self._plan_cache) > 8192:
+                    self._plan_cache.clear()
+                try:
+                    plan = self._compute_plan(state, demands, set(free), pod.name)
+                except SchedulingError:
+                    self._plan_cache[plan_key] = None
+                    raise
+                self._plan_cache[plan_key] = plan
 
-            # then wildcards over whatever remains
-            wkey = (WILDCARD, WILDCARD)
-            if wkey in demands:
-                k = len(demands[wkey])
-                cand = [state.gpus[u].index for u in sorted(free)]
-                picked = self._choose(state, cand, k)
-                if len(picked) < k:
-                    raise SchedulingError(
-                        f"node {node_name}: {len(cand)} free GPUs, pod "
-                        f"{pod.name} needs {k} more"
-                    )
-                picked_uuids = [state.index_to_uuid[i] for i in picked]
-                for (cont, req), uuid in zip(demands[wkey], picked_uuids):
+            for dkey, idx_list in plan:
+                picked_uuids = [state.index_to_uuid[i] for i in idx_list]
+                for (cont, req), uuid in zip(demands[dkey], picked_uuids):
                     bindings.append((cont, req, state.gpus[uuid].concrete_name))
                     free.discard(uuid)
                     chosen_all.append(uuid)
@@ -302,6 +303,49 @@ class GroupScheduler:
                 for uuid in chosen_all:
                     state.mark_used(uuid)
             return chosen_all
+
+    def _compute_plan(
+        self,
+        state: NodeState,
+        demands: Dict[Tuple, List],
+        free: Set[str],
+        pod_name: str,
+    ) -> List[Tuple[Tuple, List[int]]]:
+        """Index-level binding plan: ordered [(demand key, picked indices)].
+
+        Positions first (topology constraints, densest assignment via
+        _match_positions), then wildcards over the remainder.  Raises
+        SchedulingError when the node cannot satisfy the demands —
+        deterministically, so callers may cache the failure."""
+        plan: List[Tuple[Tuple, List[int]]] = []
+        pos_keys = [k for k in demands if k != (WILDCARD, WILDCARD)]
+        if pos_keys:
+            his: Dict[int, Dict[int, int]] = {}
+            for hi, gi in pos_keys:
+                his.setdefault(hi, {})[gi] = len(demands[(hi, gi)])
+            assignment = self._match_positions(state, his, free, pod_name)
+            for (hi, gi), (ah, ag) in sorted(assignment.items()):
+                members = self._group_members(state, ah, ag)
+                cand = [state.gpus[u].index for u in members if u in free]
+                k = len(demands[(hi, gi)])
+                picked = self._choose(state, cand, k)
+                for i in picked:
+                    free.discard(state.index_to_uuid[i])
+                plan.append(((hi, gi), picked))
+        wkey = (WILDCARD, WILDCARD)
+        if wkey in demands:
+            k = len(demands[wkey])
+            cand = [state.gpus[u].index for u in sorted(free)]
+            picked = self._choose(state, cand, k)
+            if len(picked) < k:
+                raise SchedulingError(
+                    f"node {state.name}: {len(cand)} free GPUs, pod "
+                    f"{pod_name} needs {k} more"
+                )
+            for i in picked:
+                free.discard(state.index_to_uuid[i])
+            plan.append((wkey, picked))
+        return plan
 
     def _match_positions(
         self,

@@ -312,3 +312,42 @@ def test_amddevs_cli_fake_modes(capsys):
     out = _json.loads(capsys.readouterr().out)
     assert len(out["gpus"]) == 4 and "/dev/kfd" in out["devices"]
     assert out["envs"]["ROCR_VISIBLE_DEVICES"].count(",") == 3
+
+
+def test_bind_plan_cache_equivalence():
+    """Plan-cache hits produce bit-identical placements to cold binds."""
+    from kubegpu_amd.discovery import FakeBackend
+
+    def build():
+        c = Cluster()
+        for n in range(2):
+            mgr = create_device_plugin(FakeBackend(fixtures.fixture_2hive_8gpu()))
+            c.add_node_from_manager(f"n{n}", mgr)
+        return c
+
+    cached, cold = build(), build()
+    seq = [2, 4, 2, 1, 8, 2, 4, 1, 2, 2, 4]
+    placements = [[], []]
+    live = [[], []]
+    for pi, (cl, out, lv, flush) in enumerate(
+        ((cached, placements[0], live[0], False), (cold, placements[1], live[1], True))
+    ):
+        for i, k in enumerate(seq * 3):
+            if flush:
+                cl.core._plan_cache.clear()  # force cold path every bind
+            pod = PodInfo(
+                name=f"p{i}",
+                running_containers={
+                    "c": ContainerInfo(kube_requests={RESOURCE_GPU: k})
+                },
+            )
+            try:
+                res = cl.schedule(pod)
+                out.append((res.node_name, tuple(sorted(res.uuids))))
+                lv.append(pod)
+            except Exception:
+                out.append(None)
+            while len(lv) > 6:
+                cl.release(lv.pop(0))
+    assert placements[0] == placements[1]
+    assert len(cached.core._plan_cache) > 0  # the cache actually engaged
