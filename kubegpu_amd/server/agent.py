@@ -37,6 +37,11 @@ def main(argv=None) -> int:
     ap.add_argument("--health-interval", type=float, default=30.0)
     ap.add_argument("--fake", action="store_true",
                     help="use the 8xMI355X fixture backend (no GPU needed)")
+    ap.add_argument("--startup-probe", action="store_true",
+                    help="run the RCCL all-reduce probe over the node's "
+                         "GPUs at start and export the measured busbw "
+                         "(the verification loop, BASELINE.json)")
+    ap.add_argument("--probe-bytes", type=int, default=256 << 20)
     ap.add_argument("--oneshot", action="store_true",
                     help="start, print state, exit (plumbing check)")
     args = ap.parse_args(argv)
@@ -47,6 +52,21 @@ def main(argv=None) -> int:
     manager = create_device_plugin(backend)
     manager.start()
     utils.logf(0, "agent: discovered %d GPU(s)", len(manager.gpus))
+
+    if args.startup_probe:
+        try:
+            from ..probe import run_rccl_probe
+
+            idxs = sorted(g.index for g in manager.gpus.values())
+            rec = run_rccl_probe(devices=idxs, nbytes=args.probe_bytes,
+                                 iters=10, warmup=3)
+            METRICS.set_xgmi_gbps(rec.get("busbw_gbps", 0.0))
+            utils.logf(
+                0, "startup probe: %d GPU(s) busbw %.1f GB/s (check=%s)",
+                len(idxs), rec.get("busbw_gbps", 0.0), rec.get("check"),
+            )
+        except Exception as e:
+            utils.errorf("startup probe failed (agent continues): %s", e)
 
     plugin = KubeletDevicePlugin(
         manager,

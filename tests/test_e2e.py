@@ -351,3 +351,28 @@ def test_bind_plan_cache_equivalence():
                 cl.release(lv.pop(0))
     assert placements[0] == placements[1]
     assert len(cached.core._plan_cache) > 0  # the cache actually engaged
+
+
+def test_agent_startup_probe_stub(tmp_path, monkeypatch, capsys):
+    """--startup-probe runs the probe binary and exports busbw; verified
+    hermetically with a stub probe."""
+    import stat
+
+    stub = tmp_path / "rcclprobe"
+    stub.write_text(
+        "#!/bin/sh\n"
+        'echo \'{"busbw_gbps": 321.5, "algbw_gbps": 321.5, "ndev": 8, "check": "pass"}\'\n'
+    )
+    stub.chmod(stub.stat().st_mode | stat.S_IEXEC)
+    monkeypatch.setenv("KUBEGPU_RCCLPROBE", str(stub))
+
+    from kubegpu_amd.metrics import METRICS
+    from kubegpu_amd.server.agent import main as agent_main
+
+    rc = agent_main([
+        "--fake", "--no-register", "--oneshot", "--startup-probe",
+        "--socket", str(tmp_path / "a.sock"), "--metrics-port", "0",
+    ])
+    assert rc == 0
+    if hasattr(METRICS, "_xgmi"):
+        assert METRICS._xgmi._value.get() == 321.5
