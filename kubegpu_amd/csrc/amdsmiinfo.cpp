@@ -55,6 +55,8 @@ struct Gpu {
   uint64_t ecc_correctable = 0;
   uint64_t ecc_uncorrectable = 0;
   uint32_t process_count = 0;
+  std::string compute_partition;  // SPX/DPX/.../CPX (MI355X partition modes)
+  std::string memory_partition;   // NPS1/NPS4/...
   std::vector<Link> links;
   amdsmi_processor_handle handle{};
 };
@@ -185,6 +187,18 @@ bool collect(std::vector<Gpu>& gpus, std::string& driver_version) {
       if (pst == AMDSMI_STATUS_SUCCESS || pst == AMDSMI_STATUS_OUT_OF_RESOURCES)
         g.process_count = nproc;
 
+      // MI355X partitioning modes (SPX..CPX / NPS1..): a partitioned
+      // GPU enumerates as multiple processors — operators need to see
+      // the mode to interpret counts and memory sizes.
+      char part[32] = {0};
+      if (amdsmi_get_gpu_compute_partition(h, part, sizeof(part)) ==
+          AMDSMI_STATUS_SUCCESS)
+        g.compute_partition = part;
+      char mpart[32] = {0};
+      if (amdsmi_get_gpu_memory_partition(h, mpart, sizeof(mpart)) ==
+          AMDSMI_STATUS_SUCCESS)
+        g.memory_partition = mpart;
+
       if (driver_version.empty()) {
         amdsmi_driver_info_t dinfo{};
         if (amdsmi_get_gpu_driver_info(h, &dinfo) == AMDSMI_STATUS_SUCCESS)
@@ -251,6 +265,8 @@ void print_json(const std::vector<Gpu>& gpus, const std::string& driver) {
     printf("   \"ecc_correctable\": %" PRIu64 ",\n", g.ecc_correctable);
     printf("   \"ecc_uncorrectable\": %" PRIu64 ",\n", g.ecc_uncorrectable);
     printf("   \"process_count\": %u,\n", g.process_count);
+    printf("   \"compute_partition\": \"%s\",\n", json_escape(g.compute_partition).c_str());
+    printf("   \"memory_partition\": \"%s\",\n", json_escape(g.memory_partition).c_str());
     printf("   \"memory\": {\"vram_total_bytes\": %" PRIu64
            ", \"vram_type\": \"%s\", \"vram_bandwidth_gbps\": %.1f},\n",
            g.vram_total_bytes, json_escape(g.vram_type).c_str(), g.vram_bandwidth_gbps);
@@ -280,6 +296,9 @@ void print_human(const std::vector<Gpu>& gpus, const std::string& driver) {
     printf("  ECC:    %" PRIu64 " correctable / %" PRIu64 " uncorrectable\n",
            g.ecc_correctable, g.ecc_uncorrectable);
     printf("  procs:  %u\n", g.process_count);
+    if (!g.compute_partition.empty() || !g.memory_partition.empty())
+      printf("  partition: compute=%s memory=%s\n", g.compute_partition.c_str(),
+             g.memory_partition.c_str());
   }
   printf("\nPairwise topology (type/hops/weight/GBps/p2p):\n");
   for (const Gpu& g : gpus) {

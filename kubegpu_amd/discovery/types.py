@@ -98,6 +98,8 @@ class GpuInfo:
     ecc_correctable: int = 0
     ecc_uncorrectable: int = 0
     process_count: int = 0  # compute processes seen by amdsmi (external in-use)
+    compute_partition: str = ""  # MI355X partition mode (SPX/DPX/.../CPX)
+    memory_partition: str = ""  # NUMA-per-socket mode (NPS1/NPS4/...)
     memory: MemoryInfo = field(default_factory=MemoryInfo)
     links: List[LinkInfo] = field(default_factory=list)
 
@@ -129,6 +131,8 @@ class GpuInfo:
             "ecc_correctable": self.ecc_correctable,
             "ecc_uncorrectable": self.ecc_uncorrectable,
             "process_count": self.process_count,
+            "compute_partition": self.compute_partition,
+            "memory_partition": self.memory_partition,
             "memory": self.memory.to_dict(),
             "links": [l.to_dict() for l in self.links],
         }
@@ -149,6 +153,8 @@ class GpuInfo:
             ecc_correctable=int(d.get("ecc_correctable", 0)),
             ecc_uncorrectable=int(d.get("ecc_uncorrectable", 0)),
             process_count=int(d.get("process_count", 0)),
+            compute_partition=str(d.get("compute_partition", "")),
+            memory_partition=str(d.get("memory_partition", "")),
             memory=MemoryInfo.from_dict(d.get("memory", {})),
             links=[LinkInfo.from_dict(x) for x in d.get("links", [])],
         )
