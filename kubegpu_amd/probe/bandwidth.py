@@ -60,7 +60,8 @@ def d2d_copy_bw_gbps(
 ) -> float:
     """Timed device-to-device streaming-copy bandwidth (GB/s, R+W).
 
-    variant: 0 = grid-stride, 1 = 4x-unrolled, 2 = contiguous-chunk.
+    variant: 0 = grid-stride NT, 1 = 4x-unrolled NT, 2 = contiguous-chunk
+    NT, 3 = regular loads + NT stores.
     """
     return load_ext().copy_bw_gbps(nbytes, iters, blocks, nontemporal, variant)
 
