@@ -88,3 +88,41 @@ def test_same_hive_whenever_possible(k):
     res = cluster.schedule(pod)
     hives = {int(u.split("-")[-1]) // 4 for u in res.uuids}
     assert len(hives) == 1
+
+
+@settings(max_examples=25, deadline=None)
+@given(
+    nodes=st.lists(st.sampled_from(sorted(FIXTURES)), min_size=1, max_size=3),
+    stream=st.lists(
+        st.tuples(st.integers(min_value=1, max_value=8), st.booleans()),
+        min_size=1,
+        max_size=25,
+    ),
+)
+def test_plan_cache_equivalence_random(nodes, stream):
+    """For any node mix and stream, the bind-plan cache never changes a
+    placement decision vs cold binds (cache cleared before every pod)."""
+    outs = []
+    for flush in (False, True):
+        cluster = _mk_cluster(nodes)
+        live = []
+        placed = []
+        for i, (k, rel) in enumerate(stream):
+            if flush:
+                cluster.core._plan_cache.clear()
+            pod = PodInfo(
+                name=f"p{i}",
+                running_containers={
+                    "c": ContainerInfo(kube_requests={RESOURCE_GPU: k})
+                },
+            )
+            try:
+                res = cluster.schedule(pod)
+                placed.append((res.node_name, tuple(sorted(res.uuids))))
+                live.append(pod)
+            except SchedulingError:
+                placed.append(None)
+            if rel and live:
+                cluster.release(live.pop(0))
+        outs.append(placed)
+    assert outs[0] == outs[1]
