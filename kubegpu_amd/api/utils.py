@@ -10,7 +10,7 @@ from __future__ import annotations
 import logging
 import os
 import sys
-from typing import Iterable, List, Mapping
+from typing import List, Mapping
 
 _logger = logging.getLogger("kubegpu_amd")
 if not _logger.handlers:

@@ -15,7 +15,6 @@ from __future__ import annotations
 
 import os
 import subprocess
-import sys
 import sysconfig
 
 PKG_DIR = os.path.dirname(os.path.abspath(__file__))

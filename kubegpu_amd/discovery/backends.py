@@ -21,9 +21,7 @@ from __future__ import annotations
 
 import abc
 import glob
-import json
 import os
-import re
 import subprocess
 from typing import Dict, List, Optional
 

@@ -14,7 +14,7 @@ from __future__ import annotations
 
 import json
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 LINK_XGMI = "XGMI"
 LINK_PCIE = "PCIE"

@@ -14,7 +14,6 @@ from __future__ import annotations
 import importlib
 import os
 import sys
-from typing import Optional
 
 _EXT_DIR = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "_ext")
 _ext = None

@@ -18,10 +18,10 @@ per-card requests, either
 from __future__ import annotations
 
 import re
-from typing import Dict, List, Optional, Tuple
+from typing import List, Tuple
 
 from ..api import utils
-from ..api.resource import WILDCARD, translate_resource
+from ..api.resource import translate_resource
 from ..api.types import (
     DEVICE_GROUP_PREFIX,
     ContainerInfo,

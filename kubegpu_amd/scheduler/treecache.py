@@ -18,12 +18,10 @@ gpu.go:168-169).  Differences from the reference, deliberate:
 
 from __future__ import annotations
 
-import re
 import threading
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Tuple
 
-from ..api import utils
 from ..api.resource import CARDS_RE
 from ..plugintypes import (
     SortedTreeNode,

@@ -21,11 +21,9 @@ manager, cf. nvidia_gpu_manager.go:132-155).
 from __future__ import annotations
 
 import os
-import queue
 import threading
-import time
 from concurrent import futures
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 import grpc
 
