@@ -129,6 +129,15 @@ __global__ void read_sum_kernel(const uint4* __restrict__ src, size_t n4,
   }
 }
 
+// Write-only streaming: NT fill (completes the read/write/copy triple).
+__global__ void fill_kernel_v4_nt(uint4v* __restrict__ dst, size_t n4,
+                                  unsigned int word) {
+  uint4v v = {word, word, word, word};
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n4; i += stride) __builtin_nontemporal_store(v, &dst[i]);
+}
+
 static void check_pair(const at::Tensor& dst, const at::Tensor& src) {
   TORCH_CHECK(src.is_cuda() && dst.is_cuda(), "gpuprobe: tensors must be on GPU");
   TORCH_CHECK(src.is_contiguous() && dst.is_contiguous(),
@@ -207,6 +216,37 @@ double copy_bw_gbps(int64_t nbytes, int64_t iters, int64_t blocks_arg,
   return (double)nbytes * 2.0 * iters / sec / 1e9;
 }
 
+double write_bw_gbps(int64_t nbytes, int64_t iters, int64_t blocks_arg) {
+  TORCH_CHECK(nbytes > 0 && nbytes % 16 == 0, "nbytes must be positive, 16-aligned");
+  auto opts = at::TensorOptions().dtype(at::kByte).device(at::kCUDA);
+  at::Tensor dst = at::empty({nbytes}, opts);
+  auto stream = at::hip::getCurrentHIPStream();
+  size_t n4 = (size_t)nbytes / 16;
+  int blocks = blocks_arg > 0
+                   ? (int)blocks_arg
+                   : (int)std::min<size_t>((n4 + BLOCK - 1) / BLOCK, DEFAULT_COPY_BLOCKS);
+  auto launch = [&]() {
+    hipLaunchKernelGGL(fill_kernel_v4_nt, dim3(blocks), dim3(BLOCK), 0, stream,
+                       (uint4v*)dst.data_ptr(), n4, 0x01010101u);
+  };
+  for (int w = 0; w < 3; ++w) launch();
+  hipEvent_t t0, t1;
+  (void)hipEventCreate(&t0);
+  (void)hipEventCreate(&t1);
+  (void)hipEventRecord(t0, stream);
+  for (int64_t i = 0; i < iters; ++i) launch();
+  (void)hipEventRecord(t1, stream);
+  (void)hipEventSynchronize(t1);
+  float ms = 0.f;
+  (void)hipEventElapsedTime(&ms, t0, t1);
+  (void)hipEventDestroy(t0);
+  (void)hipEventDestroy(t1);
+  C10_HIP_KERNEL_LAUNCH_CHECK();
+  TORCH_CHECK((int64_t)dst.sum(at::kLong).item<int64_t>() == nbytes,
+              "write_bw_gbps: fill produced wrong output");
+  return (double)nbytes * iters / (ms / 1e3) / 1e9;
+}
+
 double read_bw_gbps(int64_t nbytes, int64_t iters) {
   TORCH_CHECK(nbytes > 0 && nbytes % 16 == 0, "nbytes must be positive, 16-aligned");
   auto opts = at::TensorOptions().dtype(at::kByte).device(at::kCUDA);
@@ -246,4 +286,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("nontemporal") = true, py::arg("variant") = 0);
   m.def("read_bw_gbps", &read_bw_gbps, "timed read bandwidth",
         py::arg("nbytes"), py::arg("iters") = 20);
+  m.def("write_bw_gbps", &write_bw_gbps, "timed write-only (NT fill) bandwidth",
+        py::arg("nbytes"), py::arg("iters") = 20, py::arg("blocks") = 0);
 }

@@ -67,3 +67,8 @@ def d2d_copy_bw_gbps(
 
 def read_bw_gbps(nbytes: int = 1 << 30, iters: int = 20) -> float:
     return load_ext().read_bw_gbps(nbytes, iters)
+
+
+def write_bw_gbps(nbytes: int = 1 << 30, iters: int = 20, blocks: int = 0) -> float:
+    """Timed write-only (nontemporal fill) bandwidth (GB/s)."""
+    return load_ext().write_bw_gbps(nbytes, iters, blocks)

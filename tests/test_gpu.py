@@ -207,3 +207,17 @@ def test_amdsmiinfo_process_count_field():
     raw = json.loads(out.stdout.decode())
     assert all("process_count" in d for d in raw["devices"])
     assert all(d["process_count"] >= 0 for d in raw["devices"])
+
+
+def test_rw_bandwidth_triple():
+    """Read-only / write-only / copy bandwidths are each HBM-class and
+    mutually consistent (read and write each beat the copy R+W rate)."""
+    from kubegpu_amd.probe.bandwidth import load_ext
+
+    ext = load_ext(required=True)
+    copy = ext.copy_bw_gbps(1 << 30, 10)
+    read = ext.read_bw_gbps(1 << 30, 10)
+    write = ext.write_bw_gbps(1 << 30, 10)
+    assert read > 2000 and write > 2000 and copy > 2000
+    # one-directional streams should each exceed half the R+W copy rate
+    assert read > copy / 2 and write > copy / 2
