@@ -376,3 +376,23 @@ def test_agent_startup_probe_stub(tmp_path, monkeypatch, capsys):
     assert rc == 0
     if hasattr(METRICS, "_xgmi"):
         assert METRICS._xgmi._value.get() == 321.5
+
+
+def test_tools_smoke():
+    """Scenario scorecard + policy comparison tools stay green."""
+    import subprocess, sys, json as _json
+
+    r = subprocess.run([sys.executable, "tools/scenario_report.py"],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stdout[-500:] + r.stderr[-500:]
+    rep = _json.loads(r.stdout)
+    assert rep["all_passed"]
+
+    r = subprocess.run(
+        [sys.executable, "tools/compare_policies.py", "--pods", "200",
+         "--nodes", "2", "--topology", "mixedfleet"],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-500:]
+    out = _json.loads(r.stdout)
+    naive, ours = out["results"]
+    assert ours["pcie_bound_small_pods"] <= naive["pcie_bound_small_pods"]
