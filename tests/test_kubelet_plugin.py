@@ -314,3 +314,89 @@ def test_register_with_fake_kubelet(tmp_path):
         assert got["preferred"] is True
     finally:
         server.stop(grace=0.5)
+
+
+def test_full_kubelet_protocol_lifecycle(tmp_path):
+    """The complete kubelet<->plugin protocol in sequence: Register ->
+    GetDevicePluginOptions -> ListAndWatch -> GetPreferredAllocation ->
+    Allocate (the CPU-only 'kind cluster' flow, BASELINE.json config 1)."""
+    mgr = create_device_plugin(FakeBackend(fixtures.fixture_2hive_8gpu()))
+    mgr.start()
+    p = KubeletDevicePlugin(mgr, socket_path=str(tmp_path / "life.sock"))
+    p.start()
+    try:
+        ch = grpc.insecure_channel(f"unix://{p.socket_path}")
+        opts = _stub(ch, "GetDevicePluginOptions", dpapi.Empty,
+                     dpapi.DevicePluginOptions)(dpapi.Empty(), timeout=10)
+        assert opts.get_preferred_allocation_available
+        frame = next(iter(_stub(ch, "ListAndWatch", dpapi.Empty,
+                                dpapi.ListAndWatchResponse, streaming=True)(
+            dpapi.Empty(), timeout=10)))
+        ids = [d.ID for d in frame.devices if d.health == dpapi.HEALTHY]
+        assert len(ids) == 8
+        pref = _stub(ch, "GetPreferredAllocation",
+                     dpapi.PreferredAllocationRequest,
+                     dpapi.PreferredAllocationResponse)(
+            dpapi.PreferredAllocationRequest(container_requests=[
+                dpapi.ContainerPreferredAllocationRequest(
+                    available_deviceIDs=ids, allocation_size=2)
+            ]), timeout=10)
+        chosen = list(pref.container_responses[0].deviceIDs)
+        assert len(chosen) == 2
+        alloc = _stub(ch, "Allocate", dpapi.AllocateRequest,
+                      dpapi.AllocateResponse)(
+            dpapi.AllocateRequest(container_requests=[
+                dpapi.ContainerAllocateRequest(devicesIDs=chosen)
+            ]), timeout=10)
+        cresp = alloc.container_responses[0]
+        paths = [d.host_path for d in cresp.devices]
+        assert "/dev/kfd" in paths
+        assert sum("renderD" in x for x in paths) == 2
+        assert cresp.envs["ROCR_VISIBLE_DEVICES"] == ",".join(chosen)
+        ch.close()
+    finally:
+        p.stop()
+
+
+def test_discovery_failure_flips_unhealthy_and_recovers(tmp_path):
+    """Backend failing mid-stream: next frame goes all-Unhealthy; a
+    recovered backend flips devices back to Healthy."""
+    from kubegpu_amd.discovery import Backend, DiscoveryError
+
+    class Flaky(Backend):
+        def __init__(self, inner):
+            self.inner = inner
+            self.fail = False
+
+        def get_gpu_info(self):
+            if self.fail:
+                raise DiscoveryError("injected")
+            return self.inner.get_gpu_info()
+
+    flaky = Flaky(FakeBackend(fixtures.fixture_8x_mi355x()))
+    mgr = create_device_plugin(flaky)
+    mgr.start()
+    p = KubeletDevicePlugin(mgr, socket_path=str(tmp_path / "flaky.sock"))
+    p.start()
+    try:
+        ch = grpc.insecure_channel(f"unix://{p.socket_path}")
+        stream = _stub(ch, "ListAndWatch", dpapi.Empty,
+                       dpapi.ListAndWatchResponse, streaming=True)(
+            dpapi.Empty(), timeout=30)
+        it = iter(stream)
+        f0 = next(it)
+        assert all(d.health == dpapi.HEALTHY for d in f0.devices)
+        flaky.fail = True
+        mgr._last_get_time = 0.0  # expire the 5-min discovery cache
+        p.servicer.notify()
+        f1 = next(it)
+        assert all(d.health == dpapi.UNHEALTHY for d in f1.devices)
+        assert len(f1.devices) == 8  # devices stay visible
+        flaky.fail = False
+        mgr._last_get_time = 0.0
+        p.servicer.notify()
+        f2 = next(it)
+        assert all(d.health == dpapi.HEALTHY for d in f2.devices)
+        ch.close()
+    finally:
+        p.stop()
