@@ -96,10 +96,23 @@ def main(argv=None) -> int:
         from ..probe import probe_with_link_utilization, run_rccl_probe
 
         idxs = sorted(mgr.gpus[u].index for u in res.uuids)
+        # model prediction for the scheduled subset (the quantity the
+        # probe verifies — SURVEY.md hard part (b))
+        st = cluster.core.nodes[res.node_name]
+        pred = st.scorer.ring_bw(idxs)
+        pred = None if pred >= 1e9 else pred
         out, links = probe_with_link_utilization(
             run_rccl_probe, devices=idxs, nbytes=args.bytes
         )
-        print(json.dumps({"probe": out, "xgmi_link_traffic": links}, indent=1))
+        measured = out.get("busbw_gbps")
+        print(json.dumps({
+            "probe": out,
+            "predicted_ring_bottleneck_gbps": pred,
+            "model_vs_measured_ratio": (
+                round(measured / pred, 3) if pred and measured else None
+            ),
+            "xgmi_link_traffic": links,
+        }, indent=1))
     return 0
 
 
