@@ -396,3 +396,35 @@ def test_tools_smoke():
     out = _json.loads(r.stdout)
     naive, ours = out["results"]
     assert ours["pcie_bound_small_pods"] <= naive["pcie_bound_small_pods"]
+
+
+def test_multi_container_pod_with_init():
+    """A pod with two GPU containers + an init container binds distinct
+    GPUs per app container; the init container reuses the pod's set
+    (reference: Σ running + max init, gpu.go:296-303)."""
+    from kubegpu_amd.discovery import FakeBackend
+
+    cluster = Cluster()
+    mgr = create_device_plugin(FakeBackend(fixtures.fixture_2hive_8gpu()))
+    cluster.add_node_from_manager("n0", mgr)
+    pod = PodInfo(
+        name="multi",
+        running_containers={
+            "trainer": ContainerInfo(kube_requests={RESOURCE_GPU: 2}),
+            "loader": ContainerInfo(kube_requests={RESOURCE_GPU: 1}),
+        },
+        init_containers={"warm": ContainerInfo(kube_requests={RESOURCE_GPU: 1})},
+    )
+    res = cluster.schedule(pod)
+    assert len(res.uuids) == 3  # 2 + 1 running; init reuses
+    m_t, d_t, e_t = cluster.container_allocate(pod, "trainer")
+    m_l, d_l, e_l = cluster.container_allocate(pod, "loader")
+    m_w, d_w, e_w = cluster.container_allocate(pod, "warm")
+    t = set(e_t["ROCR_VISIBLE_DEVICES"].split(","))
+    l = set(e_l["ROCR_VISIBLE_DEVICES"].split(","))
+    w = set(e_w["ROCR_VISIBLE_DEVICES"].split(","))
+    assert len(t) == 2 and len(l) == 1 and not (t & l)
+    assert w <= (t | l)  # init ran on the pod's own set
+    # all three on one node, all with /dev/kfd
+    for d in (d_t, d_l, d_w):
+        assert d[0] == "/dev/kfd"
