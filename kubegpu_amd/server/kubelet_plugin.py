@@ -46,6 +46,7 @@ class DevicePluginServicer:
         self._stop = threading.Event()
         self._update = threading.Condition()
         self._scorer: Optional[TopologyScorer] = None
+        self._scorer_info = None
 
     # -- device list -------------------------------------------------------
 
@@ -80,8 +81,14 @@ class DevicePluginServicer:
         info = self.manager._last_info
         if info is None or not info.devices:
             return None
+        # one scorer per inventory object: topology is static between
+        # discoveries, and the scorer's memos are only valid for it
+        if self._scorer is not None and self._scorer_info is info:
+            return self._scorer
         bw = info.bandwidth_matrix()
-        return TopologyScorer([g.index for g in info.devices], bw)
+        self._scorer = TopologyScorer([g.index for g in info.devices], bw)
+        self._scorer_info = info
+        return self._scorer
 
     # -- rpc handlers ------------------------------------------------------
 
