@@ -23,6 +23,7 @@ from ..api import utils
 from ..api.types import NodeInfo, PodInfo
 from ..deviceplugin.manager import AMDGPUManager
 from ..discovery import GpusInfo
+from ..events import EVENTS
 from ..scheduler.scheduler import AMDGPUScheduler
 from ..scheduler.translate import SchedulingError
 from .group_scheduler import GroupScheduler
@@ -159,22 +160,33 @@ class Cluster:
             candidates.sort(key=lambda c: c[1])
         else:
             candidates.sort(key=lambda c: (c[0], c[1]), reverse=True)
-        _, node_name, bound_pod, uuids = candidates[0]
+        best_score, node_name, bound_pod, uuids = candidates[0]
         # adopt the winning translation/bindings into the caller's pod
         pod.running_containers = bound_pod.running_containers
         pod.init_containers = bound_pod.init_containers
         pod.node_name = node_name
         self.core.take_pod_resources(node_name, pod)
+        latency = time.perf_counter() - t0
+        ring = best_score[0] if best_score else 0.0
+        EVENTS.record(
+            "schedule",
+            pod=pod.name,
+            node=node_name,
+            gpus=list(uuids),
+            latency_ms=round(latency * 1e3, 4),
+            predicted_ring_gbps=None if ring >= 1e9 else round(ring, 1),
+        )
         return ScheduleResult(
             pod_name=pod.name,
             node_name=node_name,
             uuids=uuids,
-            latency_s=time.perf_counter() - t0,
+            latency_s=latency,
         )
 
     def release(self, pod: PodInfo) -> None:
         if pod.node_name:
             self.core.return_pod_resources(pod.node_name, pod)
+            EVENTS.record("release", pod=pod.name, node=pod.node_name)
 
     # -- container create (node side) --------------------------------------
 

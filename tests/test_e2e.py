@@ -428,3 +428,34 @@ def test_multi_container_pod_with_init():
     # all three on one node, all with /dev/kfd
     for d in (d_t, d_l, d_w):
         assert d[0] == "/dev/kfd"
+
+
+def test_event_trace_records_schedule_and_release(tmp_path, monkeypatch):
+    """EVENTS ring captures placements; KUBEGPU_EVENT_LOG appends JSONL."""
+    import json as _json
+
+    from kubegpu_amd import events
+    from kubegpu_amd.discovery import FakeBackend
+
+    log = tmp_path / "events.jsonl"
+    trace = events.EventTrace(capacity=16, path=str(log))
+    monkeypatch.setattr(events, "EVENTS", trace)
+    import kubegpu_amd.core.cluster as cluster_mod
+
+    monkeypatch.setattr(cluster_mod, "EVENTS", trace)
+
+    cluster = Cluster()
+    mgr = create_device_plugin(FakeBackend(fixtures.fixture_8x_mi355x()))
+    cluster.add_node_from_manager("n0", mgr)
+    pod = PodInfo(
+        name="traced",
+        running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: 2})},
+    )
+    cluster.schedule(pod)
+    cluster.release(pod)
+    recent = trace.recent()
+    assert [e["event"] for e in recent] == ["schedule", "release"]
+    assert recent[0]["pod"] == "traced" and len(recent[0]["gpus"]) == 2
+    assert recent[0]["predicted_ring_gbps"] >= 100
+    lines = [_json.loads(l) for l in log.read_text().splitlines()]
+    assert len(lines) == 2 and lines[0]["event"] == "schedule"
