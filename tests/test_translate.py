@@ -131,3 +131,18 @@ def test_multi_container_disjoint_slots():
     # init containers restart from slot 0 (max-over-init accounting)
     i = sorted(pod.init_containers["i"].dev_requests)
     assert i[0].endswith("/gpu/0/cards")
+
+
+def test_synth_parse_roundtrip():
+    """The synthesized request grammar and the parser are inverses."""
+    from kubegpu_amd.api.resource import parse_cards_name
+    from kubegpu_amd.scheduler.translate import synth_name
+
+    for hi in range(3):
+        for gi in range(4):
+            for card in (0, 7, 123):
+                name = synth_name(hi, gi, card)
+                prefix, h, g, cid = parse_cards_name(name)
+                assert (prefix, h, g, cid) == (
+                    "resource/group", str(hi), str(gi), str(card)
+                )
