@@ -400,3 +400,43 @@ def test_discovery_failure_flips_unhealthy_and_recovers(tmp_path):
         ch.close()
     finally:
         p.stop()
+
+
+def test_preferred_allocation_edge_cases(channel):
+    """Malformed/degenerate requests never crash the server: empty
+    request list, k=0, k > available, unknown must ids."""
+    stub = _stub(channel, "GetPreferredAllocation",
+                 dpapi.PreferredAllocationRequest,
+                 dpapi.PreferredAllocationResponse)
+    # empty request
+    resp = stub(dpapi.PreferredAllocationRequest(), timeout=10)
+    assert len(resp.container_responses) == 0
+    ids = [f"GPU-mi355x-{i:02d}" for i in range(8)]
+    # k = 0
+    resp = stub(dpapi.PreferredAllocationRequest(container_requests=[
+        dpapi.ContainerPreferredAllocationRequest(
+            available_deviceIDs=ids, allocation_size=0)]), timeout=10)
+    assert list(resp.container_responses[0].deviceIDs) == []
+    # k > available: best-effort prefix (kubelet treats as hint)
+    resp = stub(dpapi.PreferredAllocationRequest(container_requests=[
+        dpapi.ContainerPreferredAllocationRequest(
+            available_deviceIDs=ids[:2], allocation_size=5)]), timeout=10)
+    assert len(resp.container_responses[0].deviceIDs) == 2
+    # unknown must id: falls back to an unconstrained best subset
+    resp = stub(dpapi.PreferredAllocationRequest(container_requests=[
+        dpapi.ContainerPreferredAllocationRequest(
+            available_deviceIDs=ids,
+            must_include_deviceIDs=["GPU-not-real"],
+            allocation_size=2)]), timeout=10)
+    assert len(resp.container_responses[0].deviceIDs) == 2
+
+
+def test_allocate_empty_request(channel):
+    resp = _stub(channel, "Allocate", dpapi.AllocateRequest,
+                 dpapi.AllocateResponse)(
+        dpapi.AllocateRequest(container_requests=[
+            dpapi.ContainerAllocateRequest(devicesIDs=[])
+        ]), timeout=10)
+    cresp = resp.container_responses[0]
+    # /dev/kfd is still listed (harmless), but no env without GPUs
+    assert "ROCR_VISIBLE_DEVICES" not in cresp.envs
