@@ -194,9 +194,12 @@ class AMDGPUManager(Device):
                 if l.peer_index not in pos:
                     continue
                 a, b = pos[g.index], pos[l.peer_index]
-                if l.type == "XGMI" and l.hops <= 1:
+                # INTERNAL = same-package fabric between CPX/DPX
+                # partitions of one OAM: tighter than any xGMI hop, so
+                # it groups at level 0 alongside single-hop xGMI
+                if l.type in ("XGMI", "INTERNAL") and l.hops <= 1:
                     uf0.union(a, b)
-                if l.type == "XGMI":
+                if l.type in ("XGMI", "INTERNAL"):
                     uf1.union(a, b)
         for i, gi in enumerate(devs):
             for j in range(i + 1, n):

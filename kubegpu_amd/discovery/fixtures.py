@@ -143,3 +143,37 @@ def fixture_4x_no_xgmi() -> GpusInfo:
 
 def fixture_json(info: GpusInfo) -> str:
     return info.to_json()
+
+
+def fixture_cpx_2oam_8part() -> GpusInfo:
+    """Two OAMs in CPX-style partitioning, 4 compute partitions each.
+
+    Partitions of one OAM see each other over same-package INTERNAL
+    links (amdsmi link type INTERNAL); partitions on different OAMs see
+    single-hop xGMI.  Placement must prefer same-OAM subsets over
+    cross-OAM ones (INTERNAL > XGMI > PCIE).
+    """
+    from .types import INTERNAL_GBPS_DEFAULT, LINK_INTERNAL
+
+    gpus = []
+    for i in range(8):
+        g = _mk_gpu(i, numa=i // 4)
+        g.compute_partition = "CPX"
+        g.memory.vram_total_bytes = MI355X_VRAM_BYTES // 4  # per partition
+        gpus.append(g)
+    for a in gpus:
+        for b in gpus:
+            if a.index == b.index:
+                continue
+            same_oam = a.index // 4 == b.index // 4
+            if same_oam:
+                a.links.append(LinkInfo(
+                    peer_index=b.index, type=LINK_INTERNAL, hops=1,
+                    weight=5, bandwidth_gbps=INTERNAL_GBPS_DEFAULT, p2p=True,
+                ))
+            else:
+                a.links.append(LinkInfo(
+                    peer_index=b.index, type=LINK_XGMI, hops=1,
+                    weight=15, bandwidth_gbps=XGMI_LINK_GBPS_DEFAULT, p2p=True,
+                ))
+    return GpusInfo(version=_version(), devices=gpus)

@@ -18,12 +18,20 @@ from typing import Dict, List
 
 LINK_XGMI = "XGMI"
 LINK_PCIE = "PCIE"
+# Same-package fabric between compute partitions of one OAM (CPX/DPX
+# modes): amdsmi reports link type INTERNAL for them.
+LINK_INTERNAL = "INTERNAL"
 
 # Single-hop xGMI link bandwidth on MI355X (GB/s per link, spec ≈153).
 XGMI_LINK_GBPS_DEFAULT = 153.0
 # Host PCIe Gen5 x16 bandwidth (GB/s) used when two GPUs only reach each
 # other through the host bridge.
 PCIE_GBPS_DEFAULT = 63.0
+# Same-package (INTERNAL) default: partitions of one OAM share the
+# on-package Infinity Fabric / HBM path — far faster than any xGMI hop.
+# Conservative placeholder (no public per-partition figure); what
+# matters for placement is INTERNAL > XGMI > PCIE ordering.
+INTERNAL_GBPS_DEFAULT = 300.0
 
 
 @dataclass
@@ -216,12 +224,13 @@ class GpusInfo:
                     continue
                 gbps = l.bandwidth_gbps
                 if gbps <= 0.0:
-                    gbps = (
-                        XGMI_LINK_GBPS_DEFAULT / max(1, l.hops)
-                        if l.type == LINK_XGMI
-                        else PCIE_GBPS_DEFAULT
-                    )
-                if l.type == LINK_XGMI and not l.p2p:
+                    if l.type == LINK_XGMI:
+                        gbps = XGMI_LINK_GBPS_DEFAULT / max(1, l.hops)
+                    elif l.type == LINK_INTERNAL:
+                        gbps = INTERNAL_GBPS_DEFAULT
+                    else:
+                        gbps = PCIE_GBPS_DEFAULT
+                if l.type in (LINK_XGMI, LINK_INTERNAL) and not l.p2p:
                     # link advertised but peer access disabled/down:
                     # traffic bounces through the host path
                     gbps = min(gbps, PCIE_GBPS_DEFAULT)
@@ -234,10 +243,15 @@ class GpusInfo:
 
 
 def direct_xgmi_pairs(info: GpusInfo) -> List[tuple]:
-    """Pairs (i, j) connected by a single-hop xGMI link."""
+    """Pairs (i, j) connected by a single-hop xGMI or same-package
+    (INTERNAL, CPX-partition) link."""
     out = []
     for g in info.devices:
         for l in g.links:
-            if l.type == LINK_XGMI and l.hops <= 1 and g.index < l.peer_index:
+            if (
+                l.type in (LINK_XGMI, LINK_INTERNAL)
+                and l.hops <= 1
+                and g.index < l.peer_index
+            ):
                 out.append((g.index, l.peer_index))
     return out

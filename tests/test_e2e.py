@@ -459,3 +459,34 @@ def test_event_trace_records_schedule_and_release(tmp_path, monkeypatch):
     assert recent[0]["predicted_ring_gbps"] >= 100
     lines = [_json.loads(l) for l in log.read_text().splitlines()]
     assert len(lines) == 2 and lines[0]["event"] == "schedule"
+
+
+def test_cpx_partitioned_node_prefers_same_oam():
+    """CPX mode: partitions of one OAM (INTERNAL links) outrank
+    cross-OAM xGMI — a 4-partition pod gets one whole OAM, a 2-partition
+    pod never straddles OAMs."""
+    from kubegpu_amd.discovery import FakeBackend
+
+    cluster = Cluster()
+    mgr = create_device_plugin(FakeBackend(fixtures.fixture_cpx_2oam_8part()))
+    cluster.add_node_from_manager("cpx0", mgr)
+    st = cluster.core.nodes["cpx0"]
+    # bandwidth model: INTERNAL > XGMI
+    assert st.bw[0][1] > st.bw[0][4] > 100.0
+
+    pod4 = PodInfo(
+        name="oam",
+        running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: 4})},
+    )
+    res = cluster.schedule(pod4)
+    idxs = sorted(st.gpus[u].index for u in res.uuids)
+    assert idxs in ([0, 1, 2, 3], [4, 5, 6, 7])  # one intact OAM
+
+    pod2 = PodInfo(
+        name="pair",
+        running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: 2})},
+    )
+    res2 = cluster.schedule(pod2)
+    i2 = sorted(st.gpus[u].index for u in res2.uuids)
+    assert i2[0] // 4 == i2[1] // 4  # same OAM
+    assert set(i2).isdisjoint(idxs)
