@@ -213,3 +213,62 @@ def test_no_double_allocation_under_contention():
     assert not errors
     assert not overlap, f"double-allocated GPUs: {overlap[:3]}"
     assert cluster.core.free_count("n0") == 8
+
+
+def test_node_churn_during_stream():
+    """Nodes joining/leaving mid-stream: placements stay valid, released
+    state stays consistent, the tree cache's refcounts don't leak."""
+    import random
+
+    rng = random.Random(11)
+    cluster = Cluster()
+    next_node = [0]
+
+    def add_node():
+        name = f"churn{next_node[0]}"
+        next_node[0] += 1
+        fix = rng.choice(
+            [fixtures.fixture_8x_mi355x, fixtures.fixture_2hive_8gpu]
+        )()
+        mgr = create_device_plugin(FakeBackend(fix))
+        cluster.add_node_from_manager(name, mgr)
+        return name
+
+    names = [add_node() for _ in range(3)]
+    live = []
+    placed = 0
+    for i in range(300):
+        r = rng.random()
+        if r < 0.05 and len(names) < 6:
+            names.append(add_node())
+        elif r < 0.10 and len(names) > 1:
+            victim = names.pop(rng.randrange(len(names)))
+            # release pods on the victim first (kube drains before delete)
+            for pod in [p for p in live if p.node_name == victim]:
+                cluster.release(pod)
+                live.remove(pod)
+            cluster.remove_node(victim)
+        pod = PodInfo(
+            name=f"p{i}",
+            running_containers={
+                "c": ContainerInfo(
+                    kube_requests={RESOURCE_GPU: rng.choice([1, 2, 4])}
+                )
+            },
+        )
+        try:
+            res = cluster.schedule(pod)
+            assert res.node_name in names
+            live.append(pod)
+            placed += 1
+        except SchedulingError:
+            pass
+        while len(live) > 10:
+            cluster.release(live.pop(0))
+    assert placed > 200
+    # cache bookkeeping: refcounts must match live node registrations
+    cache = cluster.scheduler.cache
+    assert set(cache.node_location_map) == set(names)
+    assert sum(cache._refcount.values()) == len(names)
+    # every remaining tree key refcounted exactly
+    assert set(cache._refcount) == set(cache.node_cache_map)
