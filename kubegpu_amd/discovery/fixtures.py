@@ -10,6 +10,10 @@ nvidia_gpu_manager_test.go:16-17) with MI355X-shaped data:
   genuinely 2-level tree (gpugrp0 = hive, gpugrp1 = node).
 * ``fixture_4x_no_xgmi()``  — degenerate: no xGMI at all (each GPU its own
   group; the analog of the reference's "Topology":null K80 node).
+* ``fixture_degraded_mesh()`` — full hive with selected links DOWN
+  (pairs fall back to the host path).
+* ``fixture_cpx_2oam_8part()`` — CPX partitioning: 2 OAMs × 4 compute
+  partitions, INTERNAL links inside an OAM, xGMI across.
 """
 
 from __future__ import annotations
