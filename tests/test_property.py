@@ -126,3 +126,34 @@ def test_plan_cache_equivalence_random(nodes, stream):
                 cluster.release(live.pop(0))
         outs.append(placed)
     assert outs[0] == outs[1]
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    n=st.integers(min_value=2, max_value=8),
+    k=st.integers(min_value=1, max_value=8),
+    seed=st.integers(min_value=0, max_value=10_000),
+)
+def test_native_chooser_matches_python_random_matrices(n, k, seed):
+    """Native and Python subset choosers agree on arbitrary symmetric
+    bandwidth matrices (not just the shipped fixtures)."""
+    import random
+
+    from kubegpu_amd.scheduler.xgmi import (
+        choose_best_subset,
+        choose_best_subset_fast,
+        _native_available,
+    )
+
+    if k > n:
+        return
+    rng = random.Random(seed)
+    bw = {i: {} for i in range(n)}
+    for i in range(n):
+        for j in range(i + 1, n):
+            v = rng.choice([0.0, 32.0, 63.0, 153.0, 300.0, rng.uniform(1, 400)])
+            bw[i][j] = v
+            bw[j][i] = v
+    assert _native_available()
+    free = list(range(n))
+    assert choose_best_subset_fast(free, k, bw) == choose_best_subset(free, k, bw)
