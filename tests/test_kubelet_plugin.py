@@ -440,3 +440,32 @@ def test_allocate_empty_request(channel):
     cresp = resp.container_responses[0]
     # /dev/kfd is still listed (harmless), but no env without GPUs
     assert "ROCR_VISIBLE_DEVICES" not in cresp.envs
+
+
+def test_preferred_allocation_avoids_down_links(tmp_path):
+    """Degraded mesh through the STOCK path: GetPreferredAllocation
+    avoids pairs whose xGMI link is down (p2p=false => host-path)."""
+    fix = fixtures.fixture_degraded_mesh(missing=((0, 1), (0, 2), (0, 3), (0, 4)))
+    mgr = create_device_plugin(FakeBackend(fix))
+    mgr.start()
+    p = KubeletDevicePlugin(mgr, socket_path=str(tmp_path / "deg.sock"))
+    p.start()
+    try:
+        ch = grpc.insecure_channel(f"unix://{p.socket_path}")
+        ids = [f"GPU-mi355x-{i:02d}" for i in range(8)]
+        # ask for a pair including availability of the degraded GPU 0
+        resp = _stub(ch, "GetPreferredAllocation",
+                     dpapi.PreferredAllocationRequest,
+                     dpapi.PreferredAllocationResponse)(
+            dpapi.PreferredAllocationRequest(container_requests=[
+                dpapi.ContainerPreferredAllocationRequest(
+                    available_deviceIDs=ids, allocation_size=2)
+            ]), timeout=10)
+        pair = sorted(int(u.split("-")[-1])
+                      for u in resp.container_responses[0].deviceIDs)
+        # GPU 0 has 4 dead links; a healthy pair exists, so 0 with a dead
+        # peer must not be chosen
+        assert pair not in ([0, 1], [0, 2], [0, 3], [0, 4])
+        ch.close()
+    finally:
+        p.stop()
