@@ -34,6 +34,10 @@ def main(argv=None) -> int:
                    help="per-GPU health view (ECC totals, tombstones)")
     p.add_argument("--fake", action="store_true",
                    help="use the 8xMI355X fixture backend (no GPU needed)")
+    from .. import __version__
+
+    p.add_argument("--version", action="version",
+                   version=f"kubegpu-amd {__version__}")
     args = p.parse_args(argv)
 
     if args.fake:
