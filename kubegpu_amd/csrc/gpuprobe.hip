@@ -22,6 +22,10 @@
 // (4 per CU) + nontemporal beats larger grids — 5816 vs 4533 GB/s at
 // 4096 blocks; NT avoids LLC pollution on pure streams.
 #define DEFAULT_COPY_BLOCKS 1024
+// Pure write streams saturate earlier: >≈900 workgroups the write-drain
+// path contends and bandwidth collapses (measured: 640 wg 6,229 GB/s,
+// 1024 wg 4,300 GB/s at 1 GiB — profiles/write_bw_sweep_mi355x.json).
+#define DEFAULT_WRITE_BLOCKS 640
 
 __global__ void copy_kernel_v4(const uint4* __restrict__ src,
                                uint4* __restrict__ dst, size_t n4) {
@@ -224,7 +228,7 @@ double write_bw_gbps(int64_t nbytes, int64_t iters, int64_t blocks_arg) {
   size_t n4 = (size_t)nbytes / 16;
   int blocks = blocks_arg > 0
                    ? (int)blocks_arg
-                   : (int)std::min<size_t>((n4 + BLOCK - 1) / BLOCK, DEFAULT_COPY_BLOCKS);
+                   : (int)std::min<size_t>((n4 + BLOCK - 1) / BLOCK, DEFAULT_WRITE_BLOCKS);
   auto launch = [&]() {
     hipLaunchKernelGGL(fill_kernel_v4_nt, dim3(blocks), dim3(BLOCK), 0, stream,
                        (uint4v*)dst.data_ptr(), n4, 0x01010101u);
