@@ -251,7 +251,7 @@ double write_bw_gbps(int64_t nbytes, int64_t iters, int64_t blocks_arg) {
   return (double)nbytes * iters / (ms / 1e3) / 1e9;
 }
 
-double read_bw_gbps(int64_t nbytes, int64_t iters) {
+double read_bw_gbps(int64_t nbytes, int64_t iters, int64_t blocks_arg) {
   TORCH_CHECK(nbytes > 0 && nbytes % 16 == 0, "nbytes must be positive, 16-aligned");
   auto opts = at::TensorOptions().dtype(at::kByte).device(at::kCUDA);
   at::Tensor src = at::empty({nbytes}, opts);
@@ -259,7 +259,9 @@ double read_bw_gbps(int64_t nbytes, int64_t iters) {
   at::Tensor out = at::zeros({1}, at::TensorOptions().dtype(at::kLong).device(at::kCUDA));
   auto stream = at::hip::getCurrentHIPStream();
   size_t n4 = (size_t)nbytes / 16;
-  int blocks = (int)std::min<size_t>((n4 + BLOCK - 1) / BLOCK, 4096);
+  int blocks = blocks_arg > 0
+                   ? (int)blocks_arg
+                   : (int)std::min<size_t>((n4 + BLOCK - 1) / BLOCK, 4096);
   for (int w = 0; w < 3; ++w)
     hipLaunchKernelGGL(read_sum_kernel, dim3(blocks), dim3(BLOCK), 0, stream,
                        (const uint4*)src.data_ptr(), n4,
@@ -289,7 +291,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("nbytes"), py::arg("iters") = 20, py::arg("blocks") = 0,
         py::arg("nontemporal") = true, py::arg("variant") = 0);
   m.def("read_bw_gbps", &read_bw_gbps, "timed read bandwidth",
-        py::arg("nbytes"), py::arg("iters") = 20);
+        py::arg("nbytes"), py::arg("iters") = 20, py::arg("blocks") = 0);
   m.def("write_bw_gbps", &write_bw_gbps, "timed write-only (NT fill) bandwidth",
         py::arg("nbytes"), py::arg("iters") = 20, py::arg("blocks") = 0);
 }

@@ -65,8 +65,8 @@ def d2d_copy_bw_gbps(
     return load_ext().copy_bw_gbps(nbytes, iters, blocks, nontemporal, variant)
 
 
-def read_bw_gbps(nbytes: int = 1 << 30, iters: int = 20) -> float:
-    return load_ext().read_bw_gbps(nbytes, iters)
+def read_bw_gbps(nbytes: int = 1 << 30, iters: int = 20, blocks: int = 0) -> float:
+    return load_ext().read_bw_gbps(nbytes, iters, blocks)
 
 
 def write_bw_gbps(nbytes: int = 1 << 30, iters: int = 20, blocks: int = 0) -> float:
