@@ -49,7 +49,16 @@ class Backend(abc.ABC):
         """Return the inventory as JSON bytes (raise DiscoveryError)."""
 
     def get_devices(self) -> GpusInfo:
-        return GpusInfo.from_json(self.get_gpu_info().decode())
+        payload = self.get_gpu_info()
+        try:
+            return GpusInfo.from_json(payload.decode())
+        except (ValueError, KeyError, TypeError) as e:
+            # corrupt subprocess output must surface as a discovery
+            # failure (which Start/update tolerate), not a crash — the
+            # whole point of the subprocess isolation (SURVEY.md §5)
+            raise DiscoveryError(
+                f"unparseable inventory payload ({len(payload)} bytes): {e}"
+            ) from e
 
 
 class FakeBackend(Backend):

@@ -272,3 +272,31 @@ def test_node_churn_during_stream():
     assert sum(cache._refcount.values()) == len(names)
     # every remaining tree key refcounted exactly
     assert set(cache._refcount) == set(cache.node_cache_map)
+
+
+def test_amdsmi_backend_error_paths(tmp_path):
+    """Subprocess containment: missing binary, crashing binary, and
+    garbage output all surface as DiscoveryError (never tracebacks)."""
+    import stat
+
+    from kubegpu_amd.discovery import AmdSmiBackend, DiscoveryError, GpusInfo
+
+    with pytest.raises(DiscoveryError, match="not found"):
+        AmdSmiBackend(str(tmp_path / "nope")).get_gpu_info()
+
+    crash = tmp_path / "crash"
+    crash.write_text("#!/bin/sh\necho boom >&2\nexit 3\n")
+    crash.chmod(crash.stat().st_mode | stat.S_IEXEC)
+    with pytest.raises(DiscoveryError, match="rc=3"):
+        AmdSmiBackend(str(crash)).get_gpu_info()
+
+    garbage = tmp_path / "garbage"
+    garbage.write_text("#!/bin/sh\necho 'not json'\n")
+    garbage.chmod(garbage.stat().st_mode | stat.S_IEXEC)
+    # corrupt output: DiscoveryError (NOT a raw JSONDecodeError crash)
+    b = AmdSmiBackend(str(garbage))
+    with pytest.raises(DiscoveryError, match="unparseable"):
+        b.get_devices()
+    mgr = create_device_plugin(b)
+    mgr.start()  # must not raise (reference: Start ignores errors)
+    assert len(mgr.gpus) == 0
