@@ -18,7 +18,7 @@ nvidia_gpu_manager_test.go:16-17) with MI355X-shaped data:
 
 from __future__ import annotations
 
-from typing import List, Optional, Sequence, Set, Tuple
+from typing import List, Optional, Set, Tuple
 
 from .types import (
     LINK_PCIE,
