@@ -259,9 +259,11 @@ double read_bw_gbps(int64_t nbytes, int64_t iters, int64_t blocks_arg) {
   at::Tensor out = at::zeros({1}, at::TensorOptions().dtype(at::kLong).device(at::kCUDA));
   auto stream = at::hip::getCurrentHIPStream();
   size_t n4 = (size_t)nbytes / 16;
+  // Reads peak at 1024 wg (4 waves/CU): 6,115 GB/s vs 5,768 at 4096
+  // (profiles/read_bw_sweep_mi355x.json)
   int blocks = blocks_arg > 0
                    ? (int)blocks_arg
-                   : (int)std::min<size_t>((n4 + BLOCK - 1) / BLOCK, 4096);
+                   : (int)std::min<size_t>((n4 + BLOCK - 1) / BLOCK, DEFAULT_COPY_BLOCKS);
   for (int w = 0; w < 3; ++w)
     hipLaunchKernelGGL(read_sum_kernel, dim3(blocks), dim3(BLOCK), 0, stream,
                        (const uint4*)src.data_ptr(), n4,
