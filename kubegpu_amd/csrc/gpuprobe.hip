@@ -288,12 +288,20 @@ double read_bw_gbps(int64_t nbytes, int64_t iters, int64_t blocks_arg) {
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("copy", &copy, "streaming uint4 copy kernel (dst, src)");
+  // The timed probes release the GIL: a multi-second hipEvent-timed
+  // block must not starve other Python threads (the node agent serves
+  // gRPC from the same process — measured: with the GIL held, every
+  // concurrent RPC hit DEADLINE_EXCEEDED during a probe block).
+  m.def("copy", &copy, "streaming uint4 copy kernel (dst, src)",
+        py::call_guard<py::gil_scoped_release>());
   m.def("copy_bw_gbps", &copy_bw_gbps, "timed d2d copy bandwidth",
+        py::call_guard<py::gil_scoped_release>(),
         py::arg("nbytes"), py::arg("iters") = 20, py::arg("blocks") = 0,
         py::arg("nontemporal") = true, py::arg("variant") = 0);
   m.def("read_bw_gbps", &read_bw_gbps, "timed read bandwidth",
+        py::call_guard<py::gil_scoped_release>(),
         py::arg("nbytes"), py::arg("iters") = 20, py::arg("blocks") = 0);
   m.def("write_bw_gbps", &write_bw_gbps, "timed write-only (NT fill) bandwidth",
+        py::call_guard<py::gil_scoped_release>(),
         py::arg("nbytes"), py::arg("iters") = 20, py::arg("blocks") = 0);
 }

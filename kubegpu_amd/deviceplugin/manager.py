@@ -80,6 +80,7 @@ class AMDGPUManager(Device):
         self.index_to_id: Dict[int, str] = {}
         self._last_get_time: float = 0.0
         self._last_info: Optional[GpusInfo] = None
+        self._discovering: bool = False  # single-flight refresh guard
         # Recently-vanished GPUs kept as tombstones so the kubelet
         # device plugin can report them Unhealthy (capacity visible,
         # allocatable 0) instead of silently shrinking the node, until
@@ -110,15 +111,31 @@ class AMDGPUManager(Device):
     # -- discovery ---------------------------------------------------------
 
     def update_gpu_info(self, force: bool = False) -> None:
-        """Fetch inventory (with 5-min cache), mark/sweep, regroup."""
+        """Fetch inventory (with 5-min cache), mark/sweep, regroup.
+
+        The backend fetch (a subprocess on the amdsmi path, and slow on
+        a fully loaded GPU) runs OUTSIDE the manager lock, single-flight:
+        concurrent callers serve the previous (stale) state instead of
+        queueing behind the fetch — Allocate/GetPreferredAllocation must
+        never block tens of seconds on a refresh."""
         with self._lock:
             now = time.monotonic()
             if not force and self._last_info is not None and (
                 now - self._last_get_time < DISCOVERY_CACHE_S
             ):
                 return
+            if self._discovering:
+                return  # stale-while-revalidate
+            self._discovering = True
+        try:
             info = self._backend.get_devices()  # may raise DiscoveryError
-            self._last_get_time = now
+        except BaseException:
+            with self._lock:
+                self._discovering = False
+            raise
+        with self._lock:
+            self._discovering = False
+            self._last_get_time = time.monotonic()
             self._last_info = info
 
             # mark...

@@ -300,3 +300,44 @@ def test_amdsmi_backend_error_paths(tmp_path):
     mgr = create_device_plugin(b)
     mgr.start()  # must not raise (reference: Start ignores errors)
     assert len(mgr.gpus) == 0
+
+
+def test_discovery_single_flight_nonblocking():
+    """A slow backend fetch must not block concurrent readers: while one
+    thread discovers, others serve the previous state immediately."""
+    import time as _t
+
+    from kubegpu_amd.discovery import Backend, fixtures
+
+    class SlowBackend(Backend):
+        def __init__(self):
+            self.calls = 0
+
+        def get_gpu_info(self):
+            self.calls += 1
+            if self.calls > 1:
+                _t.sleep(1.0)  # simulated slow amdsmi under load
+            return fixtures.fixture_8x_mi355x().to_json().encode()
+
+    backend = SlowBackend()
+    mgr = create_device_plugin(backend)
+    mgr.start()  # first (fast) discovery
+    started = threading.Event()
+
+    def slow_refresh():
+        started.set()
+        mgr.update_gpu_info(force=True)
+
+    th = threading.Thread(target=slow_refresh)
+    th.start()
+    started.wait()
+    _t.sleep(0.1)  # let the slow fetch begin
+    t0 = _t.perf_counter()
+    mgr.update_gpu_info(force=True)  # must NOT wait for the slow fetch
+    dt = _t.perf_counter() - t0
+    assert dt < 0.5, f"concurrent update blocked {dt:.2f}s behind the fetch"
+    assert len(mgr.gpus) == 8  # stale state still served
+    ni = NodeInfo(name="n")
+    mgr.update_node_info(ni)  # also non-blocking path
+    th.join()
+    assert backend.calls == 2  # single-flight: no duplicate fetch
