@@ -209,3 +209,48 @@ def test_ecc_unhealthy_excluded_from_allocatable():
         pods.append(pod)
         seen.update(res.uuids)
     assert bad not in seen and len(seen) == 6
+
+
+def test_create_device_from_plugin_error_paths(tmp_path):
+    """Loader rejects missing files, modules without the factory, and
+    factories returning non-Device objects."""
+    import pytest as _pytest
+
+    from kubegpu_amd.api.device import create_device_from_plugin
+
+    with _pytest.raises(ImportError):
+        create_device_from_plugin(str(tmp_path / "missing.py"))
+
+    nofactory = tmp_path / "nofactory.py"
+    nofactory.write_text("x = 1\n")
+    with _pytest.raises(AttributeError, match="create_device_plugin"):
+        create_device_from_plugin(str(nofactory))
+
+    wrongtype = tmp_path / "wrongtype.py"
+    wrongtype.write_text("def create_device_plugin():\n    return 42\n")
+    with _pytest.raises(TypeError, match="returned"):
+        create_device_from_plugin(str(wrongtype))
+
+
+def test_container_allocate_error_paths():
+    """Unknown container name and missing node manager fail loudly."""
+    import pytest as _pytest
+
+    from kubegpu_amd.api.types import ContainerInfo, PodInfo
+    from kubegpu_amd.core import Cluster
+    from kubegpu_amd.plugintypes import RESOURCE_GPU
+    from kubegpu_amd.scheduler import SchedulingError
+
+    cluster = Cluster()
+    mgr = create_device_plugin(FakeBackend(fixtures.fixture_8x_mi355x()))
+    cluster.add_node_from_manager("n0", mgr)
+    pod = PodInfo(
+        name="p",
+        running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: 1})},
+    )
+    cluster.schedule(pod)
+    with _pytest.raises(KeyError):
+        cluster.container_allocate(pod, "nope")
+    pod.node_name = "ghost-node"
+    with _pytest.raises(SchedulingError, match="no device manager"):
+        cluster.container_allocate(pod, "c")
