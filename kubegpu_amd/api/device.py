@@ -53,7 +53,10 @@ def create_device_from_plugin(path: str) -> Device:
     ``create_device_plugin() -> Device``.
     """
     import importlib.util
+    import os
 
+    if not os.path.exists(path):
+        raise ImportError(f"device plugin path does not exist: {path}")
     spec = importlib.util.spec_from_file_location("kubegpu_amd_plugin", path)
     if spec is None or spec.loader is None:
         raise ImportError(f"cannot load device plugin from {path}")
