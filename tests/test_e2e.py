@@ -490,3 +490,29 @@ def test_cpx_partitioned_node_prefers_same_oam():
     i2 = sorted(st.gpus[u].index for u in res2.uuids)
     assert i2[0] // 4 == i2[1] // 4  # same OAM
     assert set(i2).isdisjoint(idxs)
+
+
+def test_bind_unknown_node_and_cached_infeasible():
+    """Group core error paths: unknown node; the cached-infeasible plan
+    raises identically on repeat (no silent success from the cache)."""
+    import pytest as _pytest
+
+    from kubegpu_amd.core import GroupScheduler
+    from kubegpu_amd.discovery import FakeBackend
+    from kubegpu_amd.scheduler import SchedulingError
+
+    core = GroupScheduler()
+    with _pytest.raises(SchedulingError, match="unknown node"):
+        core.bind_pod("ghost", PodInfo(name="p"))
+
+    cluster = Cluster()
+    mgr = create_device_plugin(FakeBackend(fixtures.fixture_4x_no_xgmi()))
+    cluster.add_node_from_manager("n0", mgr)
+    big = PodInfo(
+        name="big",
+        running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: 8})},
+    )
+    with _pytest.raises(SchedulingError):
+        cluster.schedule(big)  # cold path
+    with _pytest.raises(SchedulingError):
+        cluster.schedule(big)  # cached-infeasible path, same outcome
