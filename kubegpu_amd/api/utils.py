@@ -12,10 +12,27 @@ import os
 import sys
 from typing import List, Mapping
 
+class _JsonFormatter(logging.Formatter):
+    """KUBEGPU_LOG_JSON=1: one JSON object per line (log aggregators)."""
+
+    def format(self, record: logging.LogRecord) -> str:
+        import json
+
+        return json.dumps({
+            "ts": self.formatTime(record, "%Y-%m-%dT%H:%M:%S"),
+            "level": record.levelname.lower(),
+            "logger": "kubegpu_amd",
+            "msg": record.getMessage(),
+        })
+
+
 _logger = logging.getLogger("kubegpu_amd")
 if not _logger.handlers:
     _h = logging.StreamHandler(sys.stderr)
-    _h.setFormatter(logging.Formatter("%(asctime)s kubegpu_amd %(message)s"))
+    if os.environ.get("KUBEGPU_LOG_JSON"):
+        _h.setFormatter(_JsonFormatter())
+    else:
+        _h.setFormatter(logging.Formatter("%(asctime)s kubegpu_amd %(message)s"))
     _logger.addHandler(_h)
     _logger.setLevel(logging.INFO)
 
