@@ -131,7 +131,8 @@ class GroupScheduler:
         # and names only enter when a plan is applied to a node.  Value:
         # ordered [(demand key, [picked indices])], or None = infeasible.
         self._plan_cache: Dict[Tuple, Optional[List[Tuple[Tuple, List[int]]]]] = {}
-        assert policy in ("xgmi", "naive")
+        if policy not in ("xgmi", "naive"):
+            raise ValueError(f"unknown GroupScheduler policy {policy!r}")
         self.policy = policy
 
     def _choose(self, state: NodeState, cand: List[int], k: int) -> List[int]:
