@@ -377,26 +377,36 @@ def test_discovery_failure_flips_unhealthy_and_recovers(tmp_path):
     mgr = create_device_plugin(flaky)
     mgr.start()
     p = KubeletDevicePlugin(mgr, socket_path=str(tmp_path / "flaky.sock"))
+    # fast re-send cadence: gRPC prefetches the next frame eagerly, so a
+    # state change lands 1-2 frames later — poll to the expected state
+    p.servicer.health_interval_s = 0.2
     p.start()
+
+    def wait_state(it, want, frames=10):
+        last = None
+        for _ in range(frames):
+            f = next(it)
+            last = {d.ID: d.health for d in f.devices}
+            if last and all(h == want for h in last.values()):
+                return last
+        raise AssertionError(f"never reached all-{want}: {last}")
+
     try:
         ch = grpc.insecure_channel(f"unix://{p.socket_path}")
         stream = _stub(ch, "ListAndWatch", dpapi.Empty,
                        dpapi.ListAndWatchResponse, streaming=True)(
-            dpapi.Empty(), timeout=30)
+            dpapi.Empty(), timeout=60)
         it = iter(stream)
-        f0 = next(it)
-        assert all(d.health == dpapi.HEALTHY for d in f0.devices)
+        wait_state(it, dpapi.HEALTHY)
         flaky.fail = True
         mgr._last_get_time = 0.0  # expire the 5-min discovery cache
         p.servicer.notify()
-        f1 = next(it)
-        assert all(d.health == dpapi.UNHEALTHY for d in f1.devices)
-        assert len(f1.devices) == 8  # devices stay visible
+        bad = wait_state(it, dpapi.UNHEALTHY)
+        assert len(bad) == 8  # devices stay visible
         flaky.fail = False
         mgr._last_get_time = 0.0
         p.servicer.notify()
-        f2 = next(it)
-        assert all(d.health == dpapi.HEALTHY for d in f2.devices)
+        wait_state(it, dpapi.HEALTHY)
         ch.close()
     finally:
         p.stop()
