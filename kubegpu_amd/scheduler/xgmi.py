@@ -259,18 +259,24 @@ def _native_available() -> bool:
         return False
 
 
-def choose_best_subset_fast(free: Sequence[int], k: int, bw: BwMatrix) -> List[int]:
+def choose_best_subset_fast(
+    free: Sequence[int], k: int, bw: BwMatrix, must: Sequence[int] = ()
+) -> List[int]:
     """Native (C++) subset chooser when built; Python fallback otherwise."""
     if _native_available():
         from .. import _schedcore
 
         idx = sorted(set(free))
+        must_set = set(must)
+        if not must_set.issubset(idx) or len(must_set) > k:
+            return []
         n = len(idx)
+        pos = {g: i for i, g in enumerate(idx)}
         flat = [0.0] * (n * n)
         for a in range(n):
             for b in range(n):
                 if a != b:
                     flat[a * n + b] = _sym_bw(bw, idx[a], idx[b])
-        picked = _schedcore.choose_best_subset(n, k, flat)
+        picked = _schedcore.choose_best_subset(n, k, flat, [pos[g] for g in must_set])
         return [idx[p] for p in picked]
-    return choose_best_subset(free, k, bw)
+    return choose_best_subset(free, k, bw, must)
