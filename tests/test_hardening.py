@@ -341,3 +341,20 @@ def test_discovery_single_flight_nonblocking():
     mgr.update_node_info(ni)  # also non-blocking path
     th.join()
     assert backend.calls == 2  # single-flight: no duplicate fetch
+
+
+def test_json_log_mode(tmp_path):
+    """KUBEGPU_LOG_JSON=1 emits one parseable JSON object per line."""
+    import subprocess, sys
+
+    code = (
+        "from kubegpu_amd.api import utils;"
+        "utils.logf(0, 'hello %s', 'world');"
+        "utils.errorf('bad %d', 7)"
+    )
+    env = dict(os.environ, KUBEGPU_LOG_JSON="1")
+    out = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                         text=True, env=env, timeout=60)
+    lines = [json.loads(l) for l in out.stderr.strip().splitlines()]
+    assert lines[0]["msg"] == "hello world" and lines[0]["level"] == "info"
+    assert lines[1]["msg"] == "bad 7" and lines[1]["level"] == "error"
