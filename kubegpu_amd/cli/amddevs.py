@@ -88,12 +88,15 @@ def main(argv=None) -> int:
     )
     res = cluster.schedule(pod)
     mounts, devices, envs = cluster.container_allocate(pod, "c")
+    from ..events import EVENTS
+
     print(json.dumps({
         "node": res.node_name,
         "gpus": res.uuids,
         "devices": devices,
         "envs": envs,
         "schedule_latency_ms": res.latency_s * 1e3,
+        "event": (EVENTS.recent(1) or [None])[-1],
     }, indent=1))
 
     if args.probe:
