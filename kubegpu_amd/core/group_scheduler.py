@@ -281,27 +281,44 @@ class GroupScheduler:
                     free.discard(uuid)
                     chosen_all.append(uuid)
 
-            # init containers run sequentially before the app containers:
-            # bind them onto the pod's already-chosen set (first-n).
+            # Init containers run sequentially BEFORE the app containers
+            # (pod demand = max(Σ running, max init), gpu.go:295-303):
+            # each binds onto the pod's chosen set first, and when a
+            # single init container needs MORE GPUs than the app
+            # containers together, the remainder is drawn from the
+            # node's still-free GPUs.  Those extras are reserved for the
+            # pod's lifetime like the running set — stock Kubernetes
+            # reserves the effective max, so a concurrent pod can never
+            # starve a mid-flight init container.
+            init_pool: List[str] = list(chosen_all)
+            init_extras: List[str] = []
             for cont in inits:
                 reqs = [
                     r
                     for r in utils.sorted_string_keys(cont.dev_requests)
                     if r.endswith("/cards")
                 ]
-                pool = chosen_all if chosen_all else state.free_uuids()
-                if len(reqs) > len(pool):
-                    raise SchedulingError(
-                        f"init container needs {len(reqs)} GPUs, only "
-                        f"{len(pool)} available on {node_name}"
-                    )
-                for req, uuid in zip(reqs, pool):
+                if len(reqs) > len(init_pool):
+                    need = len(reqs) - len(init_pool)
+                    cand = [state.gpus[u].index for u in sorted(free)]
+                    extra = self._choose(state, cand, need)
+                    if len(extra) < need:
+                        raise SchedulingError(
+                            f"init container needs {len(reqs)} GPUs, only "
+                            f"{len(init_pool) + len(cand)} available on {node_name}"
+                        )
+                    for i in extra:
+                        uuid = state.index_to_uuid[i]
+                        free.discard(uuid)
+                        init_pool.append(uuid)
+                        init_extras.append(uuid)
+                for req, uuid in zip(reqs, init_pool):
                     bindings.append((cont, req, state.gpus[uuid].concrete_name))
 
             for cont, req, concrete in bindings:
                 cont.allocate_from[req] = concrete
             if commit:
-                for uuid in chosen_all:
+                for uuid in chosen_all + init_extras:
                     state.mark_used(uuid)
             return chosen_all
 
@@ -465,8 +482,13 @@ class GroupScheduler:
 
     @staticmethod
     def _pod_uuids(pod: PodInfo) -> Set[str]:
+        """Every GPU the pod reserves: the running containers' bound set
+        plus any init-container extras (pod demand is the effective max,
+        gpu.go:295-303, so init bindings are reserved too)."""
         uuids: Set[str] = set()
-        for cont in pod.running_containers.values():
+        for cont in list(pod.running_containers.values()) + list(
+            pod.init_containers.values()
+        ):
             for concrete in cont.allocate_from.values():
                 try:
                     _, _, _, uuid = parse_cards_name(concrete)

@@ -141,11 +141,17 @@ def translate_to_tree(
 
 
 def pod_num_gpus(pod: PodInfo) -> int:
-    """Σ over running containers + max over init containers
-    (gpu.go:296-303: init containers run sequentially)."""
+    """Pod GPU demand = max(Σ running, max init) (gpu.go:295-303).
+
+    Init containers run sequentially BEFORE the app containers start, so
+    the effective demand is the larger of the app containers' total and
+    the biggest single init container — the same effective-request math
+    stock Kubernetes uses.  (The reference sums running, then raises to
+    any larger init request; it never adds the two.)
+    """
     running = sum(set_gpu_reqs(c) for c in pod.running_containers.values())
     init = max((set_gpu_reqs(c) for c in pod.init_containers.values()), default=0)
-    return running + init
+    return max(running, init)
 
 
 def convert_to_best_gpu_requests(pod: PodInfo, cache: NodeTreeCache) -> None:
@@ -164,8 +170,8 @@ def convert_to_best_gpu_requests(pod: PodInfo, cache: NodeTreeCache) -> None:
     for name in utils.sorted_string_keys(pod.running_containers):
         offset = translate_to_tree(tree, pod.running_containers[name], slots, offset)
     # Init containers run one at a time; each starts from the pod's first
-    # slots (they may overlap each other, never the running containers'
-    # total — reference takes max over init, gpu.go:296-303).
+    # slots (they may overlap the running containers' slots and each
+    # other — demand is max(Σ running, max init), gpu.go:295-303).
     for name in utils.sorted_string_keys(pod.init_containers):
         translate_to_tree(tree, pod.init_containers[name], slots, 0)
 

@@ -404,7 +404,7 @@ def test_tools_smoke():
 def test_multi_container_pod_with_init():
     """A pod with two GPU containers + an init container binds distinct
     GPUs per app container; the init container reuses the pod's set
-    (reference: Σ running + max init, gpu.go:296-303)."""
+    (reference: pod demand = max(Σ running, max init), gpu.go:295-303)."""
     from kubegpu_amd.discovery import FakeBackend
 
     cluster = Cluster()
@@ -519,3 +519,54 @@ def test_bind_unknown_node_and_cached_infeasible():
         cluster.schedule(big)  # cold path
     with _pytest.raises(SchedulingError):
         cluster.schedule(big)  # cached-infeasible path, same outcome
+
+
+def test_8run_8init_pod_schedules_on_8gpu_node():
+    """The round-1 regression case: an 8-GPU pod with an 8-GPU init
+    container schedules on an 8-GPU node (pod demand is max(Σ running,
+    max init), /root/reference/gpuschedulerplugin/gpu.go:295-303 — the
+    init set reuses the running set entirely)."""
+    cluster = _cluster_with(("n0", fixtures.fixture_8x_mi355x()))
+    pod = PodInfo(
+        name="whole-node",
+        running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: 8})},
+        init_containers={"warm": ContainerInfo(kube_requests={RESOURCE_GPU: 8})},
+    )
+    res = cluster.schedule(pod)
+    assert len(res.uuids) == 8
+    _, _, e_c = cluster.container_allocate(pod, "c")
+    _, _, e_w = cluster.container_allocate(pod, "warm")
+    assert set(e_w["ROCR_VISIBLE_DEVICES"].split(",")) == set(
+        e_c["ROCR_VISIBLE_DEVICES"].split(",")
+    )
+
+
+def test_init_bigger_than_running_draws_extras_and_reserves_them():
+    """4 running + 6-GPU init on an 8-GPU node: schedules (demand 6),
+    the init container gets the running set plus 2 extras, and the
+    extras stay reserved for the pod's lifetime (a following 4-GPU pod
+    must NOT fit while only 2 GPUs remain unreserved) then free on
+    release."""
+    cluster = _cluster_with(("n0", fixtures.fixture_8x_mi355x()))
+    pod = PodInfo(
+        name="init-heavy",
+        running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: 4})},
+        init_containers={"i": ContainerInfo(kube_requests={RESOURCE_GPU: 6})},
+    )
+    res = cluster.schedule(pod)
+    assert len(res.uuids) == 4
+    _, _, e_c = cluster.container_allocate(pod, "c")
+    _, _, e_i = cluster.container_allocate(pod, "i")
+    run_set = set(e_c["ROCR_VISIBLE_DEVICES"].split(","))
+    init_set = set(e_i["ROCR_VISIBLE_DEVICES"].split(","))
+    assert len(init_set) == 6 and run_set <= init_set
+    # effective reservation: 8 - 6 = 2 free -> a 4-GPU pod cannot fit
+    with pytest.raises(SchedulingError):
+        cluster.schedule(_pod("too-big", 4))
+    # ...but a 2-GPU pod can
+    p2 = _pod("small", 2)
+    cluster.schedule(p2)
+    cluster.release(p2)
+    cluster.release(pod)
+    # everything freed: whole node schedulable again
+    cluster.schedule(_pod("whole", 8))

@@ -146,3 +146,48 @@ def test_synth_parse_roundtrip():
                 assert (prefix, h, g, cid) == (
                     "resource/group", str(hi), str(gi), str(card)
                 )
+
+
+def test_pod_demand_is_max_of_running_and_init():
+    """Pod GPU demand = max(Σ running, max init) — parity with
+    /root/reference/gpuschedulerplugin/gpu.go:295-303: the reference sums
+    the running containers, then raises the total to any larger single
+    init request.  It never ADDS init on top (round-1 regression)."""
+    from kubegpu_amd.scheduler.translate import pod_num_gpus
+
+    def pod(run, init):
+        return PodInfo(
+            name="p",
+            running_containers={
+                f"r{i}": ContainerInfo(kube_requests={RESOURCE_GPU: n})
+                for i, n in enumerate(run)
+            },
+            init_containers={
+                f"i{i}": ContainerInfo(kube_requests={RESOURCE_GPU: n})
+                for i, n in enumerate(init)
+            },
+        )
+
+    assert pod_num_gpus(pod([4, 4], [8])) == 8      # not 16
+    assert pod_num_gpus(pod([8], [8])) == 8         # not 16
+    assert pod_num_gpus(pod([2], [6])) == 6         # init dominates
+    assert pod_num_gpus(pod([2], [1, 6, 3])) == 6   # max over init
+    assert pod_num_gpus(pod([3], [])) == 3
+    assert pod_num_gpus(pod([], [5])) == 5
+
+
+def test_translate_8run_8init_fits_8gpu_tree():
+    """An 8-GPU pod with an 8-GPU init container translates against an
+    8-card tree (it would have demanded 16 and failed under the round-1
+    sum semantics; fits in the reference and stock Kubernetes)."""
+    cache = NodeTreeCache()
+    _cached_node(cache)  # 8-card canonical tree
+    ni = NodeInfo(name="node0")
+    pod = PodInfo(
+        name="p",
+        running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: 8})},
+        init_containers={"i": ContainerInfo(kube_requests={RESOURCE_GPU: 8})},
+    )
+    translate_pod_gpu_resources(ni, pod, cache)
+    assert len(pod.running_containers["c"].dev_requests) == 8
+    assert len(pod.init_containers["i"].dev_requests) == 8
