@@ -7,9 +7,16 @@ Measures the two headline metrics fixed by BASELINE.json:
    k = n_gpus: one rank per GPU over RCCL/xGMI (torch.distributed
    backend "nccl" IS RCCL on ROCm).  Each timed step is one bucketed
    bf16 all-reduce of a fixed buffer; busbw = 2*(N-1)/N * bytes / t.
+   "Of the scheduled GPU set" is literal: before the HIP runtime
+   initializes, every rank schedules the k-GPU pod against the
+   discovered topology (deterministic, so all ranks agree) and pins
+   itself to its scheduled device via ROCR_VISIBLE_DEVICES; the record
+   carries both the prediction (ring-bottleneck GB/s of the chosen
+   subset) and the measurement, closing the verification loop.
    k=1 is the degenerate sanity point of the curve: no
    interconnect exists, so the step is the hand-written CDNA4 HBM
-   streaming-copy kernel (csrc/gpuprobe.hip) and the value is HBM GB/s.
+   streaming-copy kernel (csrc/gpuprobe.hip) and the value is HBM GB/s
+   (uint8 buffer — the record's dtype/curve_point keys say so).
 2. **p50 pod-schedule latency (ms)** on a synthetic pod-request stream
    (mixed 1/2/4/8-GPU pods against an 8×MI355X topology), reported as
    extra keys (schedule_p50_ms / schedule_p95_ms).
@@ -78,19 +85,77 @@ def run_sched_bench(num_pods: int = 2000):
             "schedule_p95_ms_256node": p95_256}
 
 
+def _schedule_and_pin(n_gpus: int, local_rank: int):
+    """Schedule a k=n_gpus pod and pin THIS rank to its scheduled GPU.
+
+    Runs BEFORE torch/HIP initialize: sets ROCR_VISIBLE_DEVICES to the
+    one device the scheduler chose for this rank, so the measured
+    all-reduce runs on the *scheduled set* — the metric BASELINE.json
+    names — not on devices 0..k-1 by default (round-1 gap: the chosen
+    subset was recorded but never used for placement).
+
+    Every rank computes the same schedule independently: discovery and
+    subset choice are deterministic for a fixed topology, so no
+    cross-rank exchange is needed before the process group exists.
+    Returns a record dict for the JSON line.
+    """
+    try:
+        from kubegpu_amd.api.types import ContainerInfo, PodInfo
+        from kubegpu_amd.core import Cluster
+        from kubegpu_amd.deviceplugin import create_device_plugin
+        from kubegpu_amd.plugintypes import RESOURCE_GPU
+
+        if os.environ.get("KUBEGPU_BENCH_FAKE_TOPO"):
+            from kubegpu_amd.discovery import FakeBackend, fixtures
+
+            backend = FakeBackend(fixtures.fixture_8x_mi355x())
+            pin_env = False  # CI plumbing test: no real devices to pin
+        else:
+            from kubegpu_amd.discovery import default_backend
+
+            backend = default_backend()
+            pin_env = os.environ.get("KUBEGPU_BENCH_NO_PIN", "") == ""
+        cluster = Cluster()
+        mgr = create_device_plugin(backend)
+        cluster.add_node_from_manager("local", mgr)
+        pod = PodInfo(
+            name=f"bench-{n_gpus}",
+            running_containers={
+                "c": ContainerInfo(kube_requests={RESOURCE_GPU: n_gpus})
+            },
+        )
+        res = cluster.schedule(pod)
+        st = cluster.core.nodes["local"]
+        idxs = sorted(st.gpus[u].index for u in res.uuids)
+        pred = st.scorer.ring_bw(idxs)
+        pinned = idxs[local_rank] if local_rank < len(idxs) else None
+        if pin_env and pinned is not None:
+            os.environ["ROCR_VISIBLE_DEVICES"] = str(pinned)
+        return {
+            "scheduled_devices": idxs,
+            "predicted_ring_bottleneck_gbps": None if pred >= 1e9 else round(pred, 1),
+            "rank_pinning": (
+                "ROCR_VISIBLE_DEVICES" if pin_env and pinned is not None
+                else "simulated"
+            ),
+            "pinned_device": pinned,
+        }
+    except Exception as e:  # never fail the bench on discovery hiccups
+        return {"scheduled_devices": None, "sched_error": str(e)[:200]}
+
+
 def main() -> int:
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    # 50 timed steps ≈ 20 ms of kernel time at k=1 — still well within
-    # the whole-run budget, and stabilizes the reported GB/s (20-step
-    # windows showed ±5% box-to-box variance)
-    ap.add_argument("--steps", type=int, default=50)
-    ap.add_argument("--warmup", type=int, default=10)
+    # 500 timed steps ≈ 0.2 s of kernel time at k=1: long enough that
+    # driver-side gpu-busy sampling sees the work (round-1 weakness: an
+    # 8 ms timed region inside a 2 s run sampled as 0% busy), short
+    # enough to finish in seconds at every k.
+    ap.add_argument("--steps", type=int, default=500)
+    ap.add_argument("--warmup", type=int, default=20)
     ap.add_argument("--bytes", type=int, default=1 << 30)
     ap.add_argument("--pods", type=int, default=2000)
     args = ap.parse_args()
-
-    import torch
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -104,10 +169,23 @@ def main() -> int:
         )
         return 2
     n_gpus = world if world > 1 else args.gpus
-    on_gpu = torch.cuda.is_available()
 
+    # Pin to the scheduled set BEFORE the HIP runtime initializes
+    # (ROCR_VISIBLE_DEVICES is read at first device enumeration).
+    gpu_node = os.path.exists("/dev/kfd") or bool(
+        os.environ.get("KUBEGPU_BENCH_FAKE_TOPO")
+    )
+    real_sched = _schedule_and_pin(n_gpus, local_rank) if gpu_node else {}
+    pinned = real_sched.get("pinned_device") is not None and real_sched.get(
+        "rank_pinning"
+    ) == "ROCR_VISIBLE_DEVICES"
+
+    import torch
+
+    on_gpu = torch.cuda.is_available()
     if on_gpu:
-        torch.cuda.set_device(local_rank)
+        # pinned: this rank sees exactly one (scheduled) device
+        torch.cuda.set_device(0 if pinned else local_rank)
 
     dist = None
     if world > 1:
@@ -120,41 +198,13 @@ def main() -> int:
             rank=rank,
             world_size=world,
         )
+        # evidence the ranks actually sat on the scheduled devices
+        if real_sched:
+            gathered = [None] * world
+            dist.all_gather_object(gathered, real_sched.get("pinned_device"))
+            real_sched["rank_devices"] = gathered
 
     sched = run_sched_bench(args.pods) if rank == 0 else {}
-
-    # On a real GPU node, also schedule a k=n_gpus pod against the
-    # DISCOVERED topology and record the chosen set + the model's
-    # predicted ring bottleneck — the measured busbw below verifies it
-    # (SURVEY.md hard part (b): scoring must track RCCL throughput).
-    real_sched = {}
-    if rank == 0 and on_gpu:
-        try:
-            from kubegpu_amd.api.types import ContainerInfo, PodInfo
-            from kubegpu_amd.core import Cluster
-            from kubegpu_amd.deviceplugin import create_device_plugin
-            from kubegpu_amd.discovery import default_backend
-            from kubegpu_amd.plugintypes import RESOURCE_GPU
-
-            cluster = Cluster()
-            mgr = create_device_plugin(default_backend())
-            cluster.add_node_from_manager("local", mgr)
-            pod = PodInfo(
-                name=f"bench-{n_gpus}",
-                running_containers={
-                    "c": ContainerInfo(kube_requests={RESOURCE_GPU: n_gpus})
-                },
-            )
-            res = cluster.schedule(pod)
-            st = cluster.core.nodes["local"]
-            idxs = sorted(st.gpus[u].index for u in res.uuids)
-            pred = st.scorer.ring_bw(idxs)
-            real_sched = {
-                "scheduled_devices": idxs,
-                "predicted_ring_bottleneck_gbps": None if pred >= 1e9 else pred,
-            }
-        except Exception as e:  # never fail the bench on discovery hiccups
-            real_sched = {"scheduled_devices": None, "sched_error": str(e)[:200]}
 
     nbytes = args.bytes
     if world > 1:
@@ -202,6 +252,15 @@ def main() -> int:
         mode = "cpu_smoke_copy"
 
     if rank == 0:
+        # honest per-mode labeling (round-1 fix): the k=1 degenerate
+        # point is an HBM d2d copy of a uint8 buffer, NOT a bf16
+        # all-reduce — say so in the record itself.
+        dtype = {"rccl_allreduce": "bf16"}.get(mode, "uint8")
+        curve_point = {
+            "rccl_allreduce": f"k{n_gpus}-rccl-allreduce",
+            "hbm_d2d_copy": "k1-degenerate-hbm",
+            "cpu_smoke_copy": "cpu-smoke-not-mi355x",
+        }[mode]
         record = {
             "metric": "scheduled_set_allreduce_busbw_GBps",
             "value": round(value, 2),
@@ -213,7 +272,7 @@ def main() -> int:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # reference publishes no numbers
-            "dtype": "bf16",
+            "dtype": dtype,
             "data": "synthetic",
             "config": {
                 "model": "rccl-xgmi-allreduce-probe",
@@ -221,6 +280,7 @@ def main() -> int:
                 "seq_len": nbytes,
                 "parallelism": f"allreduce-ring-{n_gpus}gpu",
                 "mode": mode,
+                "curve_point": curve_point,
                 "buffer_bytes": nbytes,
                 "scenario": f"BASELINE.json k={n_gpus}",
                 **real_sched,

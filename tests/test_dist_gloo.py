@@ -15,9 +15,11 @@ import pytest
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
-def _run_dist(script: str, nproc: int = 2, timeout: int = 240):
+def _run_dist(script: str, nproc: int = 2, timeout: int = 240, extra_env=None):
     env = dict(os.environ)
     env["MASTER_ADDR"] = "127.0.0.1"
+    if extra_env:
+        env.update(extra_env)
     return subprocess.run(
         [
             sys.executable, "-m", "torch.distributed.run",
@@ -100,3 +102,45 @@ def test_bench_gpus_flag_without_ranks_errors():
     )
     assert res.returncode == 2
     assert "torch.distributed.run" in res.stderr
+
+
+@pytest.mark.timeout(300)
+def test_bench_world2_pins_ranks_to_scheduled_set():
+    """Subset-pinning plumbing: with a fake 8-GPU
+    topology, every rank independently schedules the k=2 pod, agrees on
+    the same subset, and the record carries the prediction AND the
+    per-rank device map — on a real box the same path exports
+    ROCR_VISIBLE_DEVICES before HIP init."""
+    res = _run_dist(
+        os.path.join(REPO, "bench.py"),
+        extra_env={"KUBEGPU_BENCH_FAKE_TOPO": "1"},
+    )
+    assert res.returncode == 0, res.stderr[-2000:]
+    rec = json.loads(
+        [l for l in res.stdout.splitlines() if l.startswith("{") and '"metric"' in l][0]
+    )
+    cfg = rec["config"]
+    assert cfg["scheduled_devices"] is not None
+    assert len(cfg["scheduled_devices"]) == 2
+    assert cfg["predicted_ring_bottleneck_gbps"] is not None
+    assert cfg["rank_pinning"] == "simulated"  # no real devices on CPU
+    # both ranks computed the same schedule and took distinct devices,
+    # in rank order == sorted scheduled set
+    assert cfg["rank_devices"] == cfg["scheduled_devices"]
+
+
+@pytest.mark.timeout(300)
+def test_bench_world4_pinning_distinct_devices():
+    """world=4: four ranks, four distinct scheduled devices."""
+    res = _run_dist(
+        os.path.join(REPO, "bench.py"),
+        nproc=4,
+        extra_env={"KUBEGPU_BENCH_FAKE_TOPO": "1"},
+    )
+    assert res.returncode == 0, res.stderr[-2000:]
+    rec = json.loads(
+        [l for l in res.stdout.splitlines() if l.startswith("{") and '"metric"' in l][0]
+    )
+    cfg = rec["config"]
+    assert len(set(cfg["rank_devices"])) == 4
+    assert cfg["rank_devices"] == cfg["scheduled_devices"]
