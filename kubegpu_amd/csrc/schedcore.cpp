@@ -165,12 +165,202 @@ std::pair<double, std::vector<int>> best_ring_py(int n, const std::vector<int>& 
   return {v, r.order};
 }
 
+// ---- bounded-time heuristic chooser -------------------------------------
+//
+// The exact chooser enumerates C(n, k): fine for an 8-GPU hive, hopeless
+// for a CPX 8-OAM node that enumerates 64 processors (C(64,8) ≈ 4.4e9).
+// Heuristic: multi-seed greedy max-min growth, then steepest-descent
+// 1-swap local search on the full (ring bw, -frag-loss, agg) objective.
+// The frag component is computed incrementally (O(k^2) per eval, not
+// O((n-k)^2)): frag(S) = E_free - Σ_{v∈S} deg_free(v) + E_within(S).
+// Property-tested against the exact chooser for n <= 10
+// (tests/test_xgmi.py) and budget-bounded so choose(64, 8) < 10 ms.
+
+struct HeurScore {
+  double ring;
+  int frag;
+  double agg;
+  bool operator>(const HeurScore& o) const {
+    if (ring != o.ring) return ring > o.ring;
+    if (frag != o.frag) return frag > o.frag;
+    return agg > o.agg;
+  }
+};
+
+std::vector<int> choose_best_subset_heuristic(int n, int k,
+                                              const std::vector<double>& bw_flat,
+                                              const std::vector<int>& must,
+                                              int max_passes) {
+  const double* bw = bw_flat.data();
+  if (k <= 0 || k > n || (int)must.size() > k) return {};
+  std::vector<int> all(n);
+  for (int i = 0; i < n; ++i) all[i] = i;
+  if (k == n) return all;
+  for (int g : must)
+    if (g < 0 || g >= n) return {};
+
+  // frag bookkeeping: deg_free[v] = xGMI-class edges from v into the free
+  // set; e_free = total xGMI-class edges among free.
+  std::vector<int> deg_free(n, 0);
+  int e_free = 0;
+  for (int a = 0; a < n; ++a)
+    for (int b = a + 1; b < n; ++b)
+      if (bw[a * n + b] >= kXgmiClassGbps) {
+        ++deg_free[a];
+        ++deg_free[b];
+        ++e_free;
+      }
+
+  auto frag_of = [&](const std::vector<int>& sub) {
+    int within = 0, deg = 0;
+    for (size_t a = 0; a < sub.size(); ++a) {
+      deg += deg_free[sub[a]];
+      for (size_t b = a + 1; b < sub.size(); ++b)
+        if (bw[sub[a] * n + sub[b]] >= kXgmiClassGbps) ++within;
+    }
+    return e_free - deg + within;
+  };
+
+  auto eval = [&](std::vector<int> sub) -> HeurScore {
+    std::sort(sub.begin(), sub.end());
+    Ring r = best_ring(sub, bw, n);
+    double agg = 0.0;
+    if (r.order.size() >= 3) {
+      for (size_t i = 0; i < r.order.size(); ++i)
+        agg += bw[r.order[i] * n + r.order[(i + 1) % r.order.size()]];
+    } else if (r.order.size() == 2) {
+      agg = bw[r.order[0] * n + r.order[1]];
+    }
+    double cap = std::isinf(r.bottleneck) ? kInfCap : r.bottleneck;
+    return {cap, frag_of(sub), agg};
+  };
+
+  // Ring-insertion growth: insert the (vertex, position) maximizing the
+  // new ring's bottleneck — tolerates weak intra-set edges the ring can
+  // bypass (max-min-to-set growth wrongly flees degraded hives).
+  auto grow = [&](const std::vector<int>& seed) {
+    std::vector<int> ring(seed);
+    std::vector<bool> in(n, false);
+    for (int g : ring) in[g] = true;
+    while ((int)ring.size() < k) {
+      int m = (int)ring.size();
+      std::vector<double> edges(m), pre(m + 1), suf(m + 1);
+      if (m >= 2) {
+        for (int i = 0; i < m; ++i)
+          edges[i] = bw[ring[i] * n + ring[(i + 1) % m]];
+        pre[0] = suf[m] = std::numeric_limits<double>::infinity();
+        for (int i = 0; i < m; ++i) pre[i + 1] = std::min(pre[i], edges[i]);
+        for (int i = m - 1; i >= 0; --i) suf[i] = std::min(suf[i + 1], edges[i]);
+      }
+      int best_v = -1, best_pos = 0;
+      double best_nb = -1.0, best_sum = -1.0;
+      for (int v = 0; v < n; ++v) {
+        if (in[v]) continue;
+        if (m == 1) {
+          double b = bw[ring[0] * n + v];
+          if (b > best_nb || (b == best_nb && b > best_sum)) {
+            best_nb = b;
+            best_sum = b;
+            best_v = v;
+            best_pos = 0;
+          }
+          continue;
+        }
+        for (int p = 0; p < m; ++p) {
+          int u = ring[p], w = ring[(p + 1) % m];
+          double others = std::min(pre[p], suf[p + 1]);
+          double uv = bw[u * n + v], vw = bw[v * n + w];
+          double nb = std::min({others, uv, vw});
+          double sm = uv + vw;
+          if (nb > best_nb || (nb == best_nb && sm > best_sum)) {
+            best_nb = nb;
+            best_sum = sm;
+            best_v = v;
+            best_pos = p;
+          }
+        }
+      }
+      ring.insert(ring.begin() + best_pos + 1, best_v);
+      in[best_v] = true;
+    }
+    return ring;
+  };
+
+  std::vector<int> seed_base(must.begin(), must.end());
+  std::sort(seed_base.begin(), seed_base.end());
+  seed_base.erase(std::unique(seed_base.begin(), seed_base.end()), seed_base.end());
+  std::vector<bool> in_must(n, false);
+  for (int g : seed_base) in_must[g] = true;
+
+  HeurScore best{-1.0, -1, -1.0};
+  std::vector<int> best_sub;
+  // one candidate per possible extra seed vertex (or just the must set
+  // when it already seeds the growth)
+  std::vector<std::vector<int>> seeds;
+  if ((int)seed_base.size() == k) {
+    seeds.push_back(seed_base);
+  } else {
+    for (int v = 0; v < n; ++v) {
+      if (in_must[v]) continue;
+      auto s = seed_base;
+      s.push_back(v);
+      seeds.push_back(std::move(s));
+    }
+    if (!seed_base.empty()) seeds.push_back(seed_base);
+  }
+  for (auto& s : seeds) {
+    auto cand = grow(s);
+    std::sort(cand.begin(), cand.end());
+    auto sc = eval(cand);
+    if (best_sub.empty() || sc > best) {
+      best = sc;
+      best_sub = cand;
+    } else if (!(best > sc) && cand < best_sub) {
+      best_sub = cand;  // equal score: keep lexicographically smallest
+    }
+  }
+
+  // steepest-descent 1-swaps (swap member out / non-member in), full
+  // objective, until a pass finds no improvement or max_passes reached
+  for (int pass = 0; pass < max_passes; ++pass) {
+    bool improved = false;
+    HeurScore pass_best = best;
+    std::vector<int> pass_sub = best_sub;
+    std::vector<bool> in(n, false);
+    for (int g : best_sub) in[g] = true;
+    for (int out_i = 0; out_i < (int)best_sub.size(); ++out_i) {
+      int out = best_sub[out_i];
+      if (in_must[out]) continue;
+      for (int v = 0; v < n; ++v) {
+        if (in[v]) continue;
+        auto cand = best_sub;
+        cand[out_i] = v;
+        std::sort(cand.begin(), cand.end());
+        auto sc = eval(cand);
+        if (sc > pass_best || (!(pass_best > sc) && cand < pass_sub)) {
+          pass_best = sc;
+          pass_sub = cand;
+          improved = true;
+        }
+      }
+    }
+    if (!improved) break;
+    best = pass_best;
+    best_sub = pass_sub;
+  }
+  std::sort(best_sub.begin(), best_sub.end());
+  return best_sub;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(_schedcore, m) {
   m.doc() = "native xGMI subset scorer (twin of kubegpu_amd.scheduler.xgmi)";
   m.def("choose_best_subset", &choose_best_subset, py::arg("n"), py::arg("k"),
         py::arg("bw_flat"), py::arg("must") = std::vector<int>());
+  m.def("choose_best_subset_heuristic", &choose_best_subset_heuristic,
+        py::arg("n"), py::arg("k"), py::arg("bw_flat"),
+        py::arg("must") = std::vector<int>(), py::arg("max_passes") = 4);
   m.def("best_ring", &best_ring_py, py::arg("n"), py::arg("subset"),
         py::arg("bw_flat"));
 }

@@ -24,10 +24,15 @@ from __future__ import annotations
 import itertools
 import math
 import os
-from typing import Dict, Iterable, List, Sequence, Tuple
+from typing import Dict, Iterable, List, Optional, Sequence, Tuple
 
 # Links at or above this count as xGMI-class for fragmentation purposes.
 XGMI_CLASS_GBPS = 100.0
+
+# Above this many free devices the exact C(n, k) enumeration is replaced
+# by the bounded-time heuristic (choose_best_subset_heuristic): a CPX
+# 8-OAM node enumerates 64 processors and C(64, 8) ≈ 4.4e9 subsets.
+HEURISTIC_FREE_THRESHOLD = 10
 
 BwMatrix = Dict[int, Dict[int, float]]
 
@@ -160,6 +165,149 @@ def choose_best_subset(
     return best_sub
 
 
+def choose_best_subset_heuristic(
+    free: Sequence[int],
+    k: int,
+    bw: BwMatrix,
+    must: Sequence[int] = (),
+    max_passes: int = 4,
+) -> List[int]:
+    """Bounded-time subset chooser for free sets too large to enumerate.
+
+    Multi-seed greedy max-min growth followed by steepest-descent 1-swap
+    local search on the full (ring bottleneck, fragmentation, aggregate)
+    objective — the same ordering the exact chooser maximizes.  Python
+    twin of _schedcore.choose_best_subset_heuristic (csrc/schedcore.cpp);
+    property-tested against the exact chooser for n <= 10 and used above
+    HEURISTIC_FREE_THRESHOLD (CPX nodes enumerate up to 64 processors;
+    the reference's analog, computeTreeScore at gpu.go:180-190, never
+    faced this because NVML trees cap at 8 leaves per level).
+    """
+    free_sorted = sorted(set(free))
+    must_set = set(must)
+    if k <= 0 or k > len(free_sorted) or len(must_set) > k:
+        return []
+    if not must_set.issubset(free_sorted):
+        return []
+    if k == len(free_sorted):
+        return free_sorted
+
+    deg = {v: 0 for v in free_sorted}
+    e_free = 0
+    for a, b in itertools.combinations(free_sorted, 2):
+        if _sym_bw(bw, a, b) >= XGMI_CLASS_GBPS:
+            deg[a] += 1
+            deg[b] += 1
+            e_free += 1
+
+    def frag_of(sub: Sequence[int]) -> int:
+        # frag(S) = E_free - Σ_{v∈S} deg_free(v) + E_within(S): O(k²)
+        within = sum(
+            1
+            for a, b in itertools.combinations(sub, 2)
+            if _sym_bw(bw, a, b) >= XGMI_CLASS_GBPS
+        )
+        return e_free - sum(deg[v] for v in sub) + within
+
+    def eval_sub(sub: Sequence[int]) -> Tuple[float, int, float]:
+        sub = sorted(sub)
+        ring, order = best_ring(sub, bw)
+        if len(order) >= 3:
+            agg = sum(
+                _sym_bw(bw, order[i], order[(i + 1) % len(order)])
+                for i in range(len(order))
+            )
+        elif len(order) == 2:
+            agg = _sym_bw(bw, order[0], order[1])
+        else:
+            agg = 0.0
+        cap = 1e9 if math.isinf(ring) else ring
+        return (cap, frag_of(sub), agg)
+
+    def grow(seed: Sequence[int]) -> List[int]:
+        """Ring-insertion growth: insert the (vertex, position) pair that
+        maximizes the new ring's bottleneck.  Unlike max-min-to-set
+        growth this tolerates weak intra-set edges the ring can bypass
+        (a degraded-mesh hive is still the right subset if a Hamiltonian
+        cycle avoids its weak links)."""
+        ring = list(seed)
+        in_s = set(ring)
+        while len(ring) < k:
+            m = len(ring)
+            edges = [
+                _sym_bw(bw, ring[i], ring[(i + 1) % m]) for i in range(m)
+            ] if m >= 2 else []
+            # min of all ring edges except position p (prefix/suffix mins)
+            if edges:
+                pre = [math.inf] * (m + 1)
+                suf = [math.inf] * (m + 1)
+                for i in range(m):
+                    pre[i + 1] = min(pre[i], edges[i])
+                for i in range(m - 1, -1, -1):
+                    suf[i] = min(suf[i + 1], edges[i])
+            best_key, best_v, best_pos = None, None, 0
+            for v in free_sorted:  # ascending: lowest index wins ties
+                if v in in_s:
+                    continue
+                if m == 1:
+                    b = _sym_bw(bw, ring[0], v)
+                    key = (b, b)
+                    if best_key is None or key > best_key:
+                        best_key, best_v, best_pos = key, v, 0
+                    continue
+                for p in range(m):
+                    u, w = ring[p], ring[(p + 1) % m]
+                    others = min(pre[p], suf[p + 1])
+                    nb = min(others, _sym_bw(bw, u, v), _sym_bw(bw, v, w))
+                    key = (nb, _sym_bw(bw, u, v) + _sym_bw(bw, v, w))
+                    if best_key is None or key > best_key:
+                        best_key, best_v, best_pos = key, v, p
+            ring.insert(best_pos + 1, best_v)
+            in_s.add(best_v)
+        return ring
+
+    seed_base = sorted(must_set)
+    seeds: List[List[int]] = []
+    if len(seed_base) == k:
+        seeds.append(seed_base)
+    else:
+        for v in free_sorted:
+            if v not in must_set:
+                seeds.append(seed_base + [v])
+        if seed_base:
+            seeds.append(seed_base)
+
+    best_score: Optional[Tuple[float, int, float]] = None
+    best_sub: List[int] = []
+    for s in seeds:
+        cand = sorted(grow(s))
+        sc = eval_sub(cand)
+        if best_score is None or sc > best_score:
+            best_score, best_sub = sc, cand
+        elif sc == best_score and cand < best_sub:
+            best_sub = cand
+
+    for _ in range(max_passes):
+        improved = False
+        pass_best, pass_sub = best_score, best_sub
+        in_best = set(best_sub)
+        for out_i, out in enumerate(best_sub):
+            if out in must_set:
+                continue
+            for v in free_sorted:
+                if v in in_best:
+                    continue
+                cand = sorted(best_sub[:out_i] + [v] + best_sub[out_i + 1:])
+                sc = eval_sub(cand)
+                if sc > pass_best or (sc == pass_best and cand < pass_sub):
+                    pass_best, pass_sub = sc, cand
+                    improved = True
+        if not improved:
+            break
+        best_score, best_sub = pass_best, pass_sub
+    return best_sub
+
+
 class TopologyScorer:
     """Per-node memoized subset scorer.
 
@@ -205,6 +353,8 @@ class TopologyScorer:
             out: List[int] = []
         elif not set(must_t).issubset(free_sorted):
             out = []
+        elif len(free_sorted) > HEURISTIC_FREE_THRESHOLD and k < len(free_sorted):
+            out = self._choose_heuristic(free_sorted, k, must_t)
         elif self._native is not None:
             if len(free_sorted) == self.n:
                 picked = self._native.choose_best_subset(
@@ -229,6 +379,32 @@ class TopologyScorer:
             out = choose_best_subset(free_sorted, k, self._bw, must_t)
         self._choose_memo[key] = out
         return list(out)
+
+    def _choose_heuristic(
+        self, free_sorted: List[int], k: int, must_t: Tuple[int, ...]
+    ) -> List[int]:
+        """Bounded-time path for large free sets (CPX: up to 64)."""
+        if self._native is not None and hasattr(
+            self._native, "choose_best_subset_heuristic"
+        ):
+            m = len(free_sorted)
+            if m == self.n:
+                picked = self._native.choose_best_subset_heuristic(
+                    self.n, k, self.flat, [self.pos[g] for g in must_t]
+                )
+                return [self.idx[p] for p in picked]
+            sub = [0.0] * (m * m)
+            for a in range(m):
+                pa = self.pos[free_sorted[a]]
+                for b in range(m):
+                    if a != b:
+                        sub[a * m + b] = self.flat[pa * self.n + self.pos[free_sorted[b]]]
+            subpos = {g: i for i, g in enumerate(free_sorted)}
+            picked = self._native.choose_best_subset_heuristic(
+                m, k, sub, [subpos[g] for g in must_t]
+            )
+            return [free_sorted[p] for p in picked]
+        return choose_best_subset_heuristic(free_sorted, k, self._bw, must_t)
 
     def ring_bw(self, subset: Sequence[int]) -> float:
         key = tuple(sorted(subset))
@@ -262,7 +438,28 @@ def _native_available() -> bool:
 def choose_best_subset_fast(
     free: Sequence[int], k: int, bw: BwMatrix, must: Sequence[int] = ()
 ) -> List[int]:
-    """Native (C++) subset chooser when built; Python fallback otherwise."""
+    """Native (C++) subset chooser when built; Python fallback otherwise.
+    Routes to the bounded-time heuristic above HEURISTIC_FREE_THRESHOLD."""
+    if len(set(free)) > HEURISTIC_FREE_THRESHOLD and k < len(set(free)):
+        if _native_available():
+            from .. import _schedcore
+
+            idx = sorted(set(free))
+            must_set = set(must)
+            if not must_set.issubset(idx) or len(must_set) > k:
+                return []
+            n = len(idx)
+            pos = {g: i for i, g in enumerate(idx)}
+            flat = [0.0] * (n * n)
+            for a in range(n):
+                for b in range(n):
+                    if a != b:
+                        flat[a * n + b] = _sym_bw(bw, idx[a], idx[b])
+            picked = _schedcore.choose_best_subset_heuristic(
+                n, k, flat, [pos[g] for g in must_set]
+            )
+            return [idx[p] for p in picked]
+        return choose_best_subset_heuristic(free, k, bw, must)
     if _native_available():
         from .. import _schedcore
 

@@ -152,3 +152,158 @@ def test_choose_with_must_native_matches_python(fixture_2hive):
         assert scorer.choose(free, k, must=must) == choose_best_subset(
             free, k, bw, must=must
         )
+
+
+# ---- bounded-time heuristic chooser (large free sets / CPX nodes) -------
+
+def _hive_matrix(n, hive, link=153.6, pcie=64.0, jitter_rng=None):
+    """Synthetic hive-structured symmetric bandwidth matrix."""
+    bw = {a: {} for a in range(n)}
+    for a in range(n):
+        for b in range(n):
+            if a == b:
+                continue
+            base = link if a // hive == b // hive else pcie
+            bw[a][b] = base * (jitter_rng.uniform(0.9, 1.1) if jitter_rng else 1.0)
+    for a in range(n):
+        for b in range(a + 1, n):
+            v = min(bw[a][b], bw[b][a])
+            bw[a][b] = bw[b][a] = v
+    return bw
+
+
+def test_heuristic_exact_parity_uniform_hives():
+    """On uniform hive topologies (the real MI355X shape: identical xGMI
+    links in-hive) the heuristic matches the exact chooser's score
+    exactly, for every n <= 10 and k."""
+    from kubegpu_amd.scheduler.xgmi import choose_best_subset_heuristic
+
+    for n, hive in [(8, 4), (8, 8), (9, 3), (10, 5), (10, 2)]:
+        bw = _hive_matrix(n, hive)
+        for k in range(2, min(n, 8) + 1):
+            free = list(range(n))
+            ex = choose_best_subset(free, k, bw)
+            he = choose_best_subset_heuristic(free, k, bw)
+            se = score_subset(ex, free, bw)
+            sh = score_subset(he, free, bw)
+            assert sh[0] == se[0] and sh[1] == se[1], (n, hive, k, se, sh)
+
+
+def test_heuristic_near_exact_on_degraded_jittered():
+    """Jittered + degraded meshes: heuristic ring bottleneck within 3%
+    of the exact optimum (gaps only appear under synthetic ±10% link
+    jitter; uniform hardware is exact — see the uniform test above)."""
+    import random
+
+    from kubegpu_amd.scheduler.xgmi import choose_best_subset_heuristic
+
+    worst = 1.0
+    for trial in range(25):
+        rng = random.Random(4200 + trial)
+        n = rng.choice([8, 9, 10])
+        bw = _hive_matrix(n, rng.choice([4, 5, 8]), jitter_rng=rng)
+        for _ in range(rng.randint(0, 4)):
+            a, b = rng.sample(range(n), 2)
+            v = rng.uniform(10.0, 80.0)
+            bw[a][b] = bw[b][a] = v
+        for k in (2, 3, 4, 8):
+            if k > n:
+                continue
+            free = list(range(n))
+            se = score_subset(choose_best_subset(free, k, bw), free, bw)
+            sh = score_subset(
+                choose_best_subset_heuristic(free, k, bw), free, bw
+            )
+            ratio = sh[0] / se[0] if se[0] else 1.0
+            worst = min(worst, ratio)
+            assert ratio >= 0.97, (trial, n, k, se, sh)
+    assert worst <= 1.0 + 1e-9  # heuristic can never beat the exact optimum
+
+
+def test_heuristic_native_matches_python():
+    """The C++ twin (csrc/schedcore.cpp choose_best_subset_heuristic)
+    and the Python reference return identical subsets."""
+    import random
+
+    import pytest
+
+    from kubegpu_amd.scheduler.xgmi import (
+        _sym_bw,
+        choose_best_subset_heuristic,
+    )
+
+    try:
+        from kubegpu_amd import _schedcore
+    except ImportError:
+        pytest.skip("native _schedcore not built")
+    for trial in range(10):
+        rng = random.Random(77 + trial)
+        n = rng.choice([8, 12, 16])
+        bw = _hive_matrix(n, rng.choice([4, 8]), jitter_rng=rng)
+        flat = [0.0] * (n * n)
+        for a in range(n):
+            for b in range(n):
+                if a != b:
+                    flat[a * n + b] = _sym_bw(bw, a, b)
+        for k in (2, 4, 8):
+            for must in ([], [1], [0, n - 1]):
+                he = choose_best_subset_heuristic(list(range(n)), k, bw, must)
+                hn = _schedcore.choose_best_subset_heuristic(n, k, flat, must)
+                assert he == hn, (trial, n, k, must, he, hn)
+
+
+def test_heuristic_must_include_respected():
+    from kubegpu_amd.scheduler.xgmi import choose_best_subset_heuristic
+
+    bw = _hive_matrix(16, 8)
+    picked = choose_best_subset_heuristic(list(range(16)), 4, bw, must=[9])
+    assert 9 in picked
+    assert all(8 <= g < 16 for g in picked)  # stays inside hive 1
+    # unsatisfiable constraints return []
+    assert choose_best_subset_heuristic(list(range(16)), 2, bw, must=[0, 5, 9]) == []
+    assert choose_best_subset_heuristic(list(range(8)), 2, bw, must=[12]) == []
+
+
+def test_choose_64_of_cpx_node_under_10ms():
+    """CPX reality check: a 64-processor node
+    (8 OAMs x 8 partitions) picks a whole OAM for k=8 in bounded time.
+    The 10 ms bound needs the native chooser; Python-only gets a looser
+    bound (it is the fallback, not the production path)."""
+    import time
+
+    from kubegpu_amd.scheduler.xgmi import (
+        HEURISTIC_FREE_THRESHOLD,
+        choose_best_subset_fast,
+    )
+
+    assert HEURISTIC_FREE_THRESHOLD < 64
+    bw = _hive_matrix(64, 8)
+    t0 = time.perf_counter()
+    picked = choose_best_subset_fast(list(range(64)), 8, bw)
+    elapsed_ms = (time.perf_counter() - t0) * 1e3
+    assert len(picked) == 8
+    assert len({g // 8 for g in picked}) == 1  # one whole OAM
+    try:
+        import kubegpu_amd._schedcore  # noqa: F401
+
+        # generous CI margin over the measured ~7 ms; the 10 ms claim is
+        # pinned by profiles/choose64_timing.json from a quiet box
+        assert elapsed_ms < 200, elapsed_ms
+    except ImportError:
+        assert elapsed_ms < 5000, elapsed_ms
+
+
+def test_scorer_routes_large_free_sets_to_heuristic():
+    """TopologyScorer.choose on a 64-device node completes fast and
+    picks an in-hive subset (the exact path would enumerate C(64,8))."""
+    import time
+
+    bw = _hive_matrix(64, 8)
+    scorer = TopologyScorer(list(range(64)), bw)
+    t0 = time.perf_counter()
+    picked = scorer.choose(list(range(64)), 8)
+    elapsed = time.perf_counter() - t0
+    assert len({g // 8 for g in picked}) == 1
+    assert elapsed < 5.0
+    # memoized second call
+    assert scorer.choose(list(range(64)), 8) == picked
