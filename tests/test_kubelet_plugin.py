@@ -479,3 +479,234 @@ def test_preferred_allocation_avoids_down_links(tmp_path):
         ch.close()
     finally:
         p.stop()
+
+
+def test_preferred_allocation_unsatisfiable_must_preserved(channel):
+    """Strict must_include contract (VERDICT round 1 #5): when the
+    constraint cannot be satisfied (more musts than allocation_size),
+    the musts are preserved in the response — never silently dropped."""
+    stub = _stub(channel, "GetPreferredAllocation",
+                 dpapi.PreferredAllocationRequest,
+                 dpapi.PreferredAllocationResponse)
+    ids = [f"GPU-mi355x-{i:02d}" for i in range(8)]
+    musts = ["GPU-mi355x-01", "GPU-mi355x-06", "GPU-mi355x-03"]
+    resp = stub(dpapi.PreferredAllocationRequest(container_requests=[
+        dpapi.ContainerPreferredAllocationRequest(
+            available_deviceIDs=ids,
+            must_include_deviceIDs=musts,
+            allocation_size=2)]), timeout=10)
+    got = list(resp.container_responses[0].deviceIDs)
+    assert got == musts  # declined, musts undropped
+
+
+def test_preferred_allocation_must_not_offered_is_filtered(channel):
+    """A must id absent from available_deviceIDs was never offered and
+    must not appear in the preference (ADVICE round 1 #1)."""
+    stub = _stub(channel, "GetPreferredAllocation",
+                 dpapi.PreferredAllocationRequest,
+                 dpapi.PreferredAllocationResponse)
+    avail = [f"GPU-mi355x-{i:02d}" for i in range(4)]  # hive 0 only
+    resp = stub(dpapi.PreferredAllocationRequest(container_requests=[
+        dpapi.ContainerPreferredAllocationRequest(
+            available_deviceIDs=avail,
+            must_include_deviceIDs=["GPU-mi355x-06"],  # offered? no
+            allocation_size=2)]), timeout=10)
+    got = list(resp.container_responses[0].deviceIDs)
+    assert "GPU-mi355x-06" not in got
+    assert len(got) == 2 and set(got) <= set(avail)
+    # fast path (k >= len(avail)) filters too
+    resp = stub(dpapi.PreferredAllocationRequest(container_requests=[
+        dpapi.ContainerPreferredAllocationRequest(
+            available_deviceIDs=avail,
+            must_include_deviceIDs=["GPU-mi355x-06"],
+            allocation_size=4)]), timeout=10)
+    got = list(resp.container_responses[0].deviceIDs)
+    assert "GPU-mi355x-06" not in got and set(got) == set(avail)
+
+
+def test_preferred_allocation_partial_known_musts_kept(plugin, channel):
+    """One of two musts is tombstoned (vanished): the KNOWN must is kept
+    in the chosen set instead of both being dropped (ADVICE round 1 #2)."""
+    mgr = plugin.manager
+    # simulate GPU 05 vanishing from the inventory while kubelet still
+    # offers it (stale ListAndWatch view)
+    g5 = mgr.gpus.pop("GPU-mi355x-05")
+    mgr.vanished["GPU-mi355x-05"] = (g5, 0.0)
+    try:
+        stub = _stub(channel, "GetPreferredAllocation",
+                     dpapi.PreferredAllocationRequest,
+                     dpapi.PreferredAllocationResponse)
+        ids = [f"GPU-mi355x-{i:02d}" for i in range(8)]
+        resp = stub(dpapi.PreferredAllocationRequest(container_requests=[
+            dpapi.ContainerPreferredAllocationRequest(
+                available_deviceIDs=ids,
+                must_include_deviceIDs=["GPU-mi355x-05", "GPU-mi355x-04"],
+                allocation_size=3)]), timeout=10)
+        got = list(resp.container_responses[0].deviceIDs)
+        assert "GPU-mi355x-04" in got  # the known must survives
+        assert "GPU-mi355x-05" in got  # offered must leads even if unknown
+        assert len(got) == 3
+    finally:
+        mgr.gpus["GPU-mi355x-05"] = g5
+        mgr.vanished.pop("GPU-mi355x-05", None)
+
+
+def test_preferred_allocation_prefers_idle_gpus(plugin, channel):
+    """GPUs with live allocations or external compute processes lose to
+    idle ones when enough idle devices exist (in_use surface,
+    VERDICT round 1 #6)."""
+    mgr = plugin.manager
+    # hive 0 = {00..03}, hive 1 = {04..07}; mark two hive-1 GPUs busy
+    mgr.gpus["GPU-mi355x-04"].in_use = True
+    mgr.gpus["GPU-mi355x-05"].process_count = 2
+    try:
+        stub = _stub(channel, "GetPreferredAllocation",
+                     dpapi.PreferredAllocationRequest,
+                     dpapi.PreferredAllocationResponse)
+        ids = [f"GPU-mi355x-{i:02d}" for i in range(8)]
+        resp = stub(dpapi.PreferredAllocationRequest(container_requests=[
+            dpapi.ContainerPreferredAllocationRequest(
+                available_deviceIDs=ids, allocation_size=4)]), timeout=10)
+        got = set(resp.container_responses[0].deviceIDs)
+        assert got == {f"GPU-mi355x-{i:02d}" for i in range(4)}  # idle hive
+    finally:
+        mgr.gpus["GPU-mi355x-04"].in_use = False
+        mgr.gpus["GPU-mi355x-05"].process_count = 0
+
+
+# ---- protocol fidelity: version negotiation + kubelet restart ----------
+
+class _FakeKubelet:
+    """Minimal kubelet Registration endpoint whose socket can be torn
+    down and recreated (restart simulation)."""
+
+    def __init__(self, sock_path, accept_versions=("v1beta1",)):
+        self.sock_path = sock_path
+        self.accept_versions = accept_versions
+        self.registrations = []
+        self.server = None
+
+    def start(self):
+        import threading
+        from concurrent import futures
+
+        self.event = threading.Event()
+
+        def handler(request, context):
+            if request.version not in self.accept_versions:
+                context.abort(
+                    grpc.StatusCode.INVALID_ARGUMENT,
+                    f"unsupported API version: {request.version}",
+                )
+            self.registrations.append(
+                (request.version, request.endpoint, request.resource_name)
+            )
+            self.event.set()
+            return dpapi.Empty()
+
+        if os.path.exists(self.sock_path):
+            os.unlink(self.sock_path)
+        self.server = grpc.server(futures.ThreadPoolExecutor(max_workers=2))
+        self.server.add_generic_rpc_handlers((
+            grpc.method_handlers_generic_handler(
+                dpapi.REGISTRATION_SERVICE,
+                {"Register": grpc.unary_unary_rpc_method_handler(
+                    handler,
+                    request_deserializer=dpapi.RegisterRequest.FromString,
+                    response_serializer=lambda m: m.SerializeToString(),
+                )},
+            ),
+        ))
+        self.server.add_insecure_port(f"unix://{self.sock_path}")
+        self.server.start()
+
+    def stop(self):
+        if self.server is not None:
+            self.server.stop(grace=0.2)
+            self.server = None
+        if os.path.exists(self.sock_path):
+            os.unlink(self.sock_path)
+
+
+def test_version_rejected_surfaces_registration_error(tmp_path):
+    """A kubelet that only speaks another version rejects v1beta1; the
+    plugin raises RegistrationError carrying kubelet's message instead
+    of an anonymous RpcError."""
+    from kubegpu_amd.server.kubelet_plugin import RegistrationError
+
+    kubelet = _FakeKubelet(str(tmp_path / "kubelet.sock"),
+                           accept_versions=("v1alpha2",))
+    kubelet.start()
+    try:
+        mgr = create_device_plugin(FakeBackend(fixtures.fixture_8x_mi355x()))
+        mgr.start()
+        p = KubeletDevicePlugin(mgr, socket_path=str(tmp_path / "a.sock"))
+        with pytest.raises(RegistrationError, match="unsupported API version"):
+            p.register_with_kubelet(kubelet.sock_path)
+    finally:
+        kubelet.stop()
+
+
+def test_reregister_after_kubelet_restart(tmp_path):
+    """Kubelet restart wipes its plugin registry and recreates its
+    socket; the watcher notices the new inode and re-registers, and the
+    plugin keeps serving Allocate across the restart (VERDICT #9)."""
+    import time as _time
+
+    kubelet = _FakeKubelet(str(tmp_path / "kubelet.sock"))
+    kubelet.start()
+    mgr = create_device_plugin(FakeBackend(fixtures.fixture_2hive_8gpu()))
+    mgr.start()
+    p = KubeletDevicePlugin(mgr, socket_path=str(tmp_path / "amdgpu.sock"))
+    p.start()
+    try:
+        p.register_with_kubelet(kubelet.sock_path)
+        assert len(kubelet.registrations) == 1
+        p.watch_kubelet(kubelet.sock_path, interval_s=0.1)
+
+        # kubelet "restarts": socket torn down and recreated
+        before = len(kubelet.registrations)
+        kubelet.stop()
+        _time.sleep(0.3)  # watcher sees the gap
+        kubelet.start()
+        deadline = _time.time() + 10
+        while len(kubelet.registrations) <= before and _time.time() < deadline:
+            _time.sleep(0.05)
+        assert len(kubelet.registrations) > before, "watcher did not re-register"
+
+        # plugin still serves allocations on its own (unchanged) socket
+        ch = grpc.insecure_channel(f"unix://{p.socket_path}")
+        alloc = _stub(ch, "Allocate", dpapi.AllocateRequest,
+                      dpapi.AllocateResponse)(
+            dpapi.AllocateRequest(container_requests=[
+                dpapi.ContainerAllocateRequest(
+                    devicesIDs=["GPU-mi355x-00", "GPU-mi355x-01"])
+            ]), timeout=10)
+        assert "ROCR_VISIBLE_DEVICES" in alloc.container_responses[0].envs
+        ch.close()
+    finally:
+        p.stop()
+        kubelet.stop()
+
+
+def test_plugin_socket_recreation_race(tmp_path):
+    """Restarting the plugin over a stale socket path works, and a new
+    client connects to the NEW server instance (stale-socket unlink in
+    start()); the old server's socket removal must not kill the new."""
+    mgr = create_device_plugin(FakeBackend(fixtures.fixture_8x_mi355x()))
+    mgr.start()
+    path = str(tmp_path / "same.sock")
+    p1 = KubeletDevicePlugin(mgr, socket_path=path)
+    p1.start()
+    # second instance takes over the same path (plugin upgrade pattern)
+    p2 = KubeletDevicePlugin(mgr, socket_path=path)
+    p2.start()
+    try:
+        ch = grpc.insecure_channel(f"unix://{path}")
+        opts = _stub(ch, "GetDevicePluginOptions", dpapi.Empty,
+                     dpapi.DevicePluginOptions)(dpapi.Empty(), timeout=10)
+        assert opts.get_preferred_allocation_available
+        ch.close()
+    finally:
+        p2.stop()
+        p1.stop()
