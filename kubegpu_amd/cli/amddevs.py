@@ -56,6 +56,7 @@ def main(argv=None) -> int:
             rows[uuid] = {
                 "healthy": health[uuid],
                 "present": uuid in mgr.gpus,
+                "in_use": bool(g.in_use) if g else None,
                 "ecc_correctable": g.ecc_correctable if g else None,
                 "ecc_uncorrectable": g.ecc_uncorrectable if g else None,
                 "process_count": g.process_count if g else None,

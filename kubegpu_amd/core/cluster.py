@@ -186,6 +186,14 @@ class Cluster:
     def release(self, pod: PodInfo) -> None:
         if pod.node_name:
             self.core.return_pod_resources(pod.node_name, pod)
+            # node side: clear the device plugin's in_use flags (the
+            # other half of allocate's in_use=True — no write-only state)
+            mgr = self.managers.get(pod.node_name)
+            if mgr is not None:
+                for cont in list(pod.running_containers.values()) + list(
+                    pod.init_containers.values()
+                ):
+                    mgr.release(pod, cont)
             EVENTS.record("release", pod=pod.name, node=pod.node_name)
 
     # -- container create (node side) --------------------------------------

@@ -321,6 +321,38 @@ class AMDGPUManager(Device):
                 envs["ROCR_VISIBLE_DEVICES"] = ",".join(visible)
             return [], devices, envs
 
+    def release(self, pod: PodInfo, container: ContainerInfo) -> List[str]:
+        """Inverse of allocate: clear in_use for the container's bound
+        GPUs (pod ended / binding returned).  While held, in_use still
+        survives re-discovery (reference invariant,
+        nvidia_gpu_manager.go:143-145); this is the missing other half —
+        round-1 set the flag and never cleared it.  Returns the released
+        uuids."""
+        with self._lock:
+            released: List[str] = []
+            for req_name in utils.sorted_string_keys(container.allocate_from):
+                m = ALLOCATE_RE.match(container.allocate_from[req_name])
+                if not m:
+                    continue
+                uuid = m.group(1)
+                gpu = self.gpus.get(uuid)
+                if gpu is not None and gpu.in_use:
+                    gpu.in_use = False
+                    released.append(uuid)
+            return released
+
+    def release_uuids(self, uuids: List[str]) -> None:
+        """Clear in_use for an explicit uuid list (operator tooling)."""
+        with self._lock:
+            for uuid in uuids:
+                gpu = self.gpus.get(uuid)
+                if gpu is not None:
+                    gpu.in_use = False
+
+    def in_use_uuids(self) -> List[str]:
+        with self._lock:
+            return sorted(u for u, g in self.gpus.items() if g.in_use)
+
 
 def create_device_plugin(backend: Optional[Backend] = None) -> AMDGPUManager:
     """Factory (parity: CreateDevicePlugin, plugin/nvidiagpu.go:8-10)."""

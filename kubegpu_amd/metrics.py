@@ -32,6 +32,7 @@ class Metrics:
         self.allocations = 0
         self.failures = 0
         self.gpu_health: Dict[str, bool] = {}
+        self.gpu_in_use: Dict[str, bool] = {}
         if _HAVE_PROM:
             # per-instance registry: multiple Metrics objects (tests,
             # embedded schedulers) must not collide in the global one
@@ -66,6 +67,12 @@ class Metrics:
                 ["uuid"],
                 registry=self.registry,
             )
+            self._gpu_in_use = Gauge(
+                "kubegpu_amd_gpu_in_use",
+                "1 = GPU held by a live allocation (manager path)",
+                ["uuid"],
+                registry=self.registry,
+            )
 
     def observe_schedule(self, seconds: float) -> None:
         with self._lock:
@@ -95,6 +102,12 @@ class Metrics:
         if _HAVE_PROM:
             self._gpu_health.labels(uuid=uuid).set(1.0 if healthy else 0.0)
             self._gpu_ecc.labels(uuid=uuid).set(float(ecc_uncorrectable))
+
+    def set_gpu_in_use(self, uuid: str, in_use: bool) -> None:
+        with self._lock:
+            self.gpu_in_use[uuid] = in_use
+        if _HAVE_PROM:
+            self._gpu_in_use.labels(uuid=uuid).set(1.0 if in_use else 0.0)
 
     def percentile(self, q: float) -> Optional[float]:
         with self._lock:

@@ -114,6 +114,9 @@ def main(argv=None) -> int:
             METRICS.set_gpu_health(
                 uuid, ok, g.ecc_uncorrectable if g is not None else 0
             )
+            METRICS.set_gpu_in_use(
+                uuid, bool(g.in_use) if g is not None else False
+            )
         time.sleep(args.health_interval if registered else 5.0)
 
     plugin.stop()

@@ -178,7 +178,15 @@ class DevicePluginServicer:
         return [*unknown_musts, *(idx_to_uuid[i] for i in picked)]
 
     def allocate(self, request, context):
-        """Device IDs -> DeviceSpecs + env (SURVEY.md §3.3 analog)."""
+        """Device IDs -> DeviceSpecs + env (SURVEY.md §3.3 analog).
+
+        Deliberately does NOT set GpuInfo.in_use: the v1beta1 API has no
+        deallocate RPC, so a plugin-side flag could never be cleared
+        (write-only state, round-1 VERDICT #6).  Kubelet owns allocation
+        accounting on this path; _is_idle uses amdsmi's process_count
+        to see actual occupancy, and in_use stays the KubeDevice-path
+        manager's (allocate/release-paired) flag.
+        """
         out = []
         for creq in request.container_requests:
             specs = [
@@ -201,7 +209,6 @@ class DevicePluginServicer:
                             )
                         )
                 visible.append(uuid)
-                g.in_use = True
             resp = dpapi.ContainerAllocateResponse(devices=specs)
             if visible:
                 resp.envs["ROCR_VISIBLE_DEVICES"] = ",".join(visible)

@@ -254,3 +254,69 @@ def test_container_allocate_error_paths():
     pod.node_name = "ghost-node"
     with _pytest.raises(SchedulingError, match="no device manager"):
         cluster.container_allocate(pod, "c")
+
+
+def test_in_use_full_lifecycle():
+    """in_use is no longer write-only: allocate
+    sets it, it survives re-discovery while held (reference invariant
+    nvidia_gpu_manager.go:143-145), release clears it, and
+    in_use_uuids() surfaces it."""
+    from kubegpu_amd.api.types import ContainerInfo, PodInfo
+    from kubegpu_amd.core import Cluster
+    from kubegpu_amd.plugintypes import RESOURCE_GPU
+
+    mgr = create_device_plugin(FakeBackend(fixtures.fixture_8x_mi355x()))
+    cluster = Cluster()
+    cluster.add_node_from_manager("n0", mgr)
+    pod = PodInfo(
+        name="p",
+        running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: 2})},
+    )
+    res = cluster.schedule(pod)
+    cluster.container_allocate(pod, "c")
+    assert mgr.in_use_uuids() == sorted(res.uuids)
+
+    # survives a forced re-discovery
+    mgr.update_gpu_info(force=True)
+    assert mgr.in_use_uuids() == sorted(res.uuids)
+
+    cluster.release(pod)
+    assert mgr.in_use_uuids() == []
+
+
+def test_release_uuids_direct():
+    mgr = create_device_plugin(FakeBackend(fixtures.fixture_8x_mi355x()))
+    mgr.start()
+    uuids = sorted(mgr.gpus)[:3]
+    for u in uuids:
+        mgr.gpus[u].in_use = True
+    mgr.release_uuids(uuids[:2] + ["GPU-not-real"])
+    assert mgr.in_use_uuids() == [uuids[2]]
+
+
+def test_kubelet_allocate_does_not_touch_in_use(tmp_path):
+    """The v1beta1 path has no deallocate RPC, so the kubelet server
+    must not set a flag it can never clear; occupancy there comes from
+    amdsmi process_count."""
+    import grpc
+
+    from kubegpu_amd.server import KubeletDevicePlugin, dpapi
+
+    mgr = create_device_plugin(FakeBackend(fixtures.fixture_8x_mi355x()))
+    mgr.start()
+    p = KubeletDevicePlugin(mgr, socket_path=str(tmp_path / "k.sock"))
+    p.start()
+    try:
+        ch = grpc.insecure_channel(f"unix://{p.socket_path}")
+        alloc = ch.unary_unary(
+            f"/{dpapi.DEVICE_PLUGIN_SERVICE}/Allocate",
+            request_serializer=lambda m: m.SerializeToString(),
+            response_deserializer=dpapi.AllocateResponse.FromString,
+        )
+        alloc(dpapi.AllocateRequest(container_requests=[
+            dpapi.ContainerAllocateRequest(devicesIDs=["GPU-mi355x-00"])
+        ]), timeout=10)
+        assert mgr.in_use_uuids() == []
+        ch.close()
+    finally:
+        p.stop()
