@@ -111,7 +111,9 @@ def test_allocate_device_specs(channel, plugin):
     assert "/dev/kfd" in paths
     assert "/dev/dri/renderD129" in paths and "/dev/dri/renderD130" in paths
     assert cr.envs["ROCR_VISIBLE_DEVICES"] == "GPU-mi355x-01,GPU-mi355x-02"
-    assert plugin.manager.gpus["GPU-mi355x-01"].in_use
+    # v1beta1 has no deallocate RPC, so the kubelet path must NOT set a
+    # flag it could never clear
+    assert not plugin.manager.gpus["GPU-mi355x-01"].in_use
 
 
 def test_allocate_unknown_device_rejected(channel):
