@@ -178,6 +178,13 @@ class GroupScheduler:
             tuple(tuple(sorted(state.gpus[u].index for u in ids)) for _, ids in g_items)
             for _, g_items in layout.groups
         )
+        # Topology fingerprint interns bandwidths at 1e-3 GB/s — a
+        # DELIBERATE tolerance: nodes whose matrices
+        # differ only below 1 MB/s share a topo_token, bind-plan cache
+        # entries and dedup classes.  Real xGMI links differ by whole
+        # GB/s (link width/count), and sub-1e-3 noise should not defeat
+        # the cache; at worst a tie-break lands on an equivalently
+        # scored subset of a node whose links differ immeasurably.
         bw_fp = tuple(
             sorted(
                 (i, j, round(v, 3))

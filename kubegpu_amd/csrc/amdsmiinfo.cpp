@@ -245,9 +245,39 @@ bool collect(std::vector<Gpu>& gpus, std::string& driver_version) {
   return true;
 }
 
+// ROCm stack version, queried at runtime (not hardcoded): prefer the
+// installed tree's /opt/rocm/.info/version, fall back to "unknown".
+std::string rocm_version() {
+  FILE* f = fopen("/opt/rocm/.info/version", "r");
+  if (f) {
+    char buf[64] = {0};
+    if (fgets(buf, sizeof(buf), f)) {
+      fclose(f);
+      std::string s(buf);
+      while (!s.empty() && (s.back() == '\n' || s.back() == '\r')) s.pop_back();
+      if (!s.empty()) return s;
+    } else {
+      fclose(f);
+    }
+  }
+  return "unknown";
+}
+
+// amdsmi library version via amdsmi_get_lib_version
+std::string amdsmi_lib_version() {
+  amdsmi_version_t v{};
+  if (amdsmi_get_lib_version(&v) == AMDSMI_STATUS_SUCCESS) {
+    char buf[48];
+    snprintf(buf, sizeof(buf), "%u.%u.%u", v.major, v.minor, v.release);
+    return buf;
+  }
+  return "lib";
+}
+
 void print_json(const std::vector<Gpu>& gpus, const std::string& driver) {
-  printf("{\n \"version\": {\"driver\": \"%s\", \"rocm\": \"7.2.0\", \"amdsmi\": \"lib\"},\n",
-         json_escape(driver).c_str());
+  printf("{\n \"version\": {\"driver\": \"%s\", \"rocm\": \"%s\", \"amdsmi\": \"%s\"},\n",
+         json_escape(driver).c_str(), json_escape(rocm_version()).c_str(),
+         json_escape(amdsmi_lib_version()).c_str());
   printf(" \"devices\": [\n");
   for (size_t i = 0; i < gpus.size(); ++i) {
     const Gpu& g = gpus[i];

@@ -81,6 +81,8 @@ class AMDGPUManager(Device):
         self._last_get_time: float = 0.0
         self._last_info: Optional[GpusInfo] = None
         self._discovering: bool = False  # single-flight refresh guard
+        self._fetch_cv = threading.Condition(self._lock)
+        self._fetch_gen = 0  # bumped on every completed fetch
         # Recently-vanished GPUs kept as tombstones so the kubelet
         # device plugin can report them Unhealthy (capacity visible,
         # allocatable 0) instead of silently shrinking the node, until
@@ -125,17 +127,33 @@ class AMDGPUManager(Device):
             ):
                 return
             if self._discovering:
-                return  # stale-while-revalidate
+                if not force:
+                    return  # stale-while-revalidate
+                # force=True must actually refresh: wait
+                # for the in-flight fetch; if it succeeded, that IS the
+                # refresh — otherwise fall through and fetch ourselves.
+                start_gen = self._fetch_gen
+                while self._discovering:
+                    self._fetch_cv.wait(timeout=30.0)
+                if self._fetch_gen > start_gen and self._last_info is not None:
+                    return
             self._discovering = True
         try:
             info = self._backend.get_devices()  # may raise DiscoveryError
         except BaseException:
             with self._lock:
                 self._discovering = False
+                self._fetch_cv.notify_all()
             raise
         with self._lock:
             self._discovering = False
-            self._last_get_time = time.monotonic()
+            self._fetch_gen += 1
+            self._fetch_cv.notify_all()
+            # re-read the clock: the fetch above may be slow, and the
+            # tombstone sweep below must stamp vanish time at NOW, not
+            # at fetch start (otherwise VANISHED_TTL shrinks)
+            now = time.monotonic()
+            self._last_get_time = now
             self._last_info = info
 
             # mark...

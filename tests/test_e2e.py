@@ -460,6 +460,7 @@ def test_event_trace_records_schedule_and_release(tmp_path, monkeypatch):
     assert [e["event"] for e in recent] == ["schedule", "release"]
     assert recent[0]["pod"] == "traced" and len(recent[0]["gpus"]) == 2
     assert recent[0]["predicted_ring_gbps"] >= 100
+    trace.flush()  # file writes are async (off the scheduling lock)
     lines = [_json.loads(l) for l in log.read_text().splitlines()]
     assert len(lines) == 2 and lines[0]["event"] == "schedule"
 

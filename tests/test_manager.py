@@ -320,3 +320,59 @@ def test_kubelet_allocate_does_not_touch_in_use(tmp_path):
         ch.close()
     finally:
         p.stop()
+
+
+def test_forced_refresh_waits_for_inflight_fetch():
+    """force=True must not silently no-op while another thread is
+    mid-fetch: it waits for the in-flight fetch and
+    returns with REFRESHED state."""
+    import threading
+    import time as _time
+
+    from kubegpu_amd.discovery import Backend
+
+    class SlowBackend(Backend):
+        def __init__(self, inner):
+            self.inner = inner
+            self.calls = 0
+            self.release = threading.Event()
+
+        def get_gpu_info(self):
+            return self.inner.get_gpu_info()
+
+        def get_devices(self):
+            self.calls += 1
+            if self.calls == 2:
+                self.release.wait(10)  # second (forced) fetch is slow
+            return self.inner.get_devices()
+
+    slow = SlowBackend(FakeBackend(fixtures.fixture_8x_mi355x()))
+    mgr = create_device_plugin(slow)
+    mgr.start()  # fetch #1
+
+    started = threading.Event()
+
+    def slow_force():
+        started.set()
+        mgr.update_gpu_info(force=True)  # fetch #2, blocks on release
+
+    t = threading.Thread(target=slow_force)
+    t.start()
+    started.wait(5)
+    _time.sleep(0.1)  # let the thread enter the backend call
+    done = {}
+
+    def second_force():
+        mgr.update_gpu_info(force=True)  # must WAIT, then see fresh state
+        done["gen"] = mgr._fetch_gen
+
+    t2 = threading.Thread(target=second_force)
+    t2.start()
+    _time.sleep(0.2)
+    assert t2.is_alive()  # waiting, not no-oping
+    slow.release.set()
+    t.join(10)
+    t2.join(10)
+    assert not t2.is_alive()
+    assert done["gen"] >= 2  # observed the completed refresh
+    assert slow.calls == 2  # waiter reused the in-flight fetch
