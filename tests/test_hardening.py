@@ -333,7 +333,12 @@ def test_discovery_single_flight_nonblocking():
     started.wait()
     _t.sleep(0.1)  # let the slow fetch begin
     t0 = _t.perf_counter()
-    mgr.update_gpu_info(force=True)  # must NOT wait for the slow fetch
+    # non-forced refresh must NOT wait for the slow fetch: Allocate /
+    # GetPreferredAllocation call this and serve stale-while-revalidate
+    # (force=True DOES wait now — ADVICE r1 #4; see
+    # test_forced_refresh_waits_for_inflight_fetch)
+    mgr._last_get_time = 0.0  # expire the cache so the path is exercised
+    mgr.update_gpu_info()
     dt = _t.perf_counter() - t0
     assert dt < 0.5, f"concurrent update blocked {dt:.2f}s behind the fetch"
     assert len(mgr.gpus) == 8  # stale state still served
