@@ -94,11 +94,17 @@ def main(argv=None) -> int:
         plugin.stop()
         return 0
 
+    watcher_started = False
     while not stop["flag"]:
         if not args.no_register and not registered:
             try:
                 plugin.register_with_kubelet(args.kubelet_socket)
                 registered = True
+                if not watcher_started:
+                    # kubelet-side restart detection: its socket is
+                    # recreated on restart; the watcher re-registers
+                    plugin.watch_kubelet(args.kubelet_socket)
+                    watcher_started = True
             except Exception as e:
                 utils.logf(2, "agent: kubelet registration pending: %s", e)
         # kubelet restart detection: our socket vanishes when the plugin
