@@ -30,7 +30,10 @@ PCIE_GBPS_DEFAULT = 63.0
 # Same-package (INTERNAL) default: partitions of one OAM share the
 # on-package Infinity Fabric / HBM path — far faster than any xGMI hop.
 # Conservative placeholder (no public per-partition figure); what
-# matters for placement is INTERNAL > XGMI > PCIE ordering.
+# matters for placement is INTERNAL > XGMI > PCIE ordering.  A measured
+# replacement needs a CPX-partitioned lease: the round-2 box reported
+# SPX/NPS1 only (profiles/partition_modes_mi355x.json), so this stays a
+# documented placeholder, never reported as a measurement.
 INTERNAL_GBPS_DEFAULT = 300.0
 
 
