@@ -129,6 +129,14 @@ def _schedule_and_pin(n_gpus: int, local_rank: int):
         idxs = sorted(st.gpus[u].index for u in res.uuids)
         pred = st.scorer.ring_bw(idxs)
         pinned = idxs[local_rank] if local_rank < len(idxs) else None
+        # amdsmi index -> BDF, so the rank can later verify the device
+        # the HIP runtime actually handed it IS the scheduled one
+        # (ROCR_VISIBLE_DEVICES indices are ROCr enumeration order,
+        # which normally matches amdsmi's BDF order — this check makes
+        # a mismatch visible in the record instead of silent)
+        idx_to_bdf = {}
+        for g in (mgr._last_info.devices if mgr._last_info else []):
+            idx_to_bdf[g.index] = g.bdf
         if pin_env and pinned is not None:
             os.environ["ROCR_VISIBLE_DEVICES"] = str(pinned)
         return {
@@ -139,6 +147,7 @@ def _schedule_and_pin(n_gpus: int, local_rank: int):
                 else "simulated"
             ),
             "pinned_device": pinned,
+            "pinned_bdf": idx_to_bdf.get(pinned),
         }
     except Exception as e:  # never fail the bench on discovery hiccups
         return {"scheduled_devices": None, "sched_error": str(e)[:200]}
@@ -186,6 +195,22 @@ def main() -> int:
     if on_gpu:
         # pinned: this rank sees exactly one (scheduled) device
         torch.cuda.set_device(0 if pinned else local_rank)
+        if pinned and real_sched.get("pinned_bdf"):
+            # cross-check: the device HIP handed us must be the one the
+            # scheduler chose (guards amdsmi-vs-ROCr enumeration skew)
+            try:
+                props = torch.cuda.get_device_properties(0)
+                bus = getattr(props, "pci_bus_id", None)
+                dom = getattr(props, "pci_domain_id", 0) or 0
+                dev = getattr(props, "pci_device_id", None)
+                if bus is not None and dev is not None:
+                    actual = f"{dom:04x}:{bus:02x}:{dev:02x}.0"
+                    want = real_sched["pinned_bdf"].lower()
+                    real_sched["bdf_verified"] = actual.lower() == want
+                    if not real_sched["bdf_verified"]:
+                        real_sched["actual_bdf"] = actual
+            except Exception:
+                pass
 
     dist = None
     if world > 1:

@@ -157,3 +157,64 @@ def test_native_chooser_matches_python_random_matrices(n, k, seed):
     assert _native_available()
     free = list(range(n))
     assert choose_best_subset_fast(free, k, bw) == choose_best_subset(free, k, bw)
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    stream=st.lists(
+        st.tuples(
+            st.integers(min_value=0, max_value=6),   # Σ running request
+            st.integers(min_value=0, max_value=8),   # max init request
+            st.booleans(),                           # release later?
+        ),
+        min_size=1,
+        max_size=25,
+    ),
+)
+def test_init_demand_invariants(stream):
+    """Pods with init containers (demand = max(Σ running, max init),
+    gpu.go:295-303): for any schedule/release stream
+    - every init container binds exactly its requested GPU count
+    - the running set is a subset of the init set when init >= running
+    - reserved GPUs == union of running + init bindings, and release
+      restores them all."""
+    cluster = _mk_cluster(["dense"])
+    node = next(iter(cluster.core.nodes))
+    total = len(cluster.core.nodes[node].gpus)
+    live = []
+    for i, (run, init, rel) in enumerate(stream):
+        if run == 0 and init == 0:
+            continue
+        running = (
+            {"c": ContainerInfo(kube_requests={RESOURCE_GPU: run})} if run else {}
+        )
+        inits = (
+            {"i": ContainerInfo(kube_requests={RESOURCE_GPU: init})} if init else {}
+        )
+        pod = PodInfo(name=f"p{i}", running_containers=running,
+                      init_containers=inits)
+        try:
+            res = cluster.schedule(pod)
+        except SchedulingError:
+            demand = max(run, init)
+            free = total - sum(d for (_, d) in live)
+            assert demand > free  # only fails when it truly cannot fit
+            continue
+        assert len(res.uuids) == run
+        bound_init = set()
+        if init:
+            cont = pod.init_containers["i"]
+            assert len(cont.allocate_from) == init
+            bound_init = {v.split("/gpu/")[1].split("/")[0]
+                          for v in cont.allocate_from.values()}
+            assert set(res.uuids) <= bound_init or init < run
+        demand = len(set(res.uuids) | bound_init)
+        assert demand == max(run, init)
+        if rel:
+            cluster.release(pod)
+        else:
+            live.append((pod, demand))
+    for pod, _ in live:
+        cluster.release(pod)
+    st_node = cluster.core.nodes[node]
+    assert len(st_node.free_uuids()) == total  # everything restored
