@@ -221,3 +221,44 @@ def test_rw_bandwidth_triple():
     assert read > 2000 and write > 2000 and copy > 2000
     # one-directional streams should each exceed half the R+W copy rate
     assert read > copy / 2 and write > copy / 2
+
+
+def test_bench_record_is_pinned_and_self_describing():
+    """Round-2 bench contract on hardware: the rank pins itself to the
+    scheduled device before HIP init (rank_pinning), the record is
+    honest about the k=1 degenerate mode (dtype/curve_point), and the
+    pinned device's BDF matches what the scheduler chose."""
+    import subprocess as _sp
+    import sys as _sys
+
+    res = _sp.run(
+        [_sys.executable, os.path.join(REPO, "bench.py"),
+         "--steps", "20", "--warmup", "5", "--pods", "50"],
+        capture_output=True, timeout=240, cwd=REPO, text=True,
+    )
+    assert res.returncode == 0, res.stderr[-1500:]
+    rec = json.loads(res.stdout.strip().splitlines()[-1])
+    cfg = rec["config"]
+    assert rec["dtype"] == "uint8"
+    assert cfg["curve_point"] == "k1-degenerate-hbm"
+    assert cfg["rank_pinning"] == "ROCR_VISIBLE_DEVICES"
+    assert cfg["scheduled_devices"] == [cfg["pinned_device"]]
+    assert cfg.get("bdf_verified") is True, cfg
+    assert rec["value"] > 3000  # HBM copy sanity floor
+
+
+def test_amddevs_health_shows_in_use():
+    """--health surfaces the allocation lifecycle field on a real node."""
+    import subprocess as _sp
+    import sys as _sys
+
+    res = _sp.run(
+        [_sys.executable, "-m", "kubegpu_amd.cli.amddevs", "--health"],
+        capture_output=True, timeout=120, cwd=REPO, text=True,
+    )
+    assert res.returncode == 0, res.stderr[-1000:]
+    rows = json.loads(res.stdout)
+    assert rows
+    for row in rows.values():
+        assert row["in_use"] is False  # fresh box: nothing allocated
+        assert "process_count" in row
