@@ -54,6 +54,20 @@ class Cluster:
         self.policy = policy
         self.node_infos: Dict[str, NodeInfo] = {}
         self.managers: Dict[str, AMDGPUManager] = {}
+        # Incremental equivalence classes: signature -> node names.  A
+        # node's signature only changes at the events this class owns
+        # (bind commit, release, node add/remove), so per-pod work is
+        # O(distinct states), not O(nodes) — flat out to multi-thousand
+        #-node fleets (docs/ROADMAP.md #5).
+        from sortedcontainers import SortedSet
+
+        self._sorted_set = SortedSet
+        self._node_sig: Dict[str, Tuple] = {}
+        # SortedSet members: O(log n) add/discard, O(1) min/max, so the
+        # per-class representative (the member the full candidate sort
+        # would pick) costs nothing to maintain even when scheduling
+        # always consumes the current representative
+        self._classes: Dict[Tuple, "SortedSet"] = {}
 
     # -- cluster state -----------------------------------------------------
 
@@ -69,6 +83,39 @@ class Cluster:
         self.node_infos[name] = node_info
         if manager is not None:
             self.managers[name] = manager
+        self.reindex_node(name)
+
+    def reindex_node(self, name: str) -> None:
+        """Re-home *name* in the signature classes after a state change.
+
+        Called automatically on add/remove/bind/release; call it
+        manually only after mutating a node's core state out-of-band.
+        """
+        new_sig = self.core.state_signature(name)
+        if new_sig is None:
+            new_sig = ("__unregistered__", name)
+        old = self._node_sig.get(name)
+        if old == new_sig:
+            return
+        if old is not None:
+            members = self._classes.get(old)
+            if members is not None:
+                members.discard(name)
+                if not members:
+                    del self._classes[old]
+        if new_sig not in self._classes:
+            self._classes[new_sig] = self._sorted_set()
+        self._classes[new_sig].add(name)
+        self._node_sig[name] = new_sig
+
+    def _drop_node_index(self, name: str) -> None:
+        old = self._node_sig.pop(name, None)
+        if old is not None:
+            members = self._classes.get(old)
+            if members is not None:
+                members.discard(name)
+                if not members:
+                    del self._classes[old]
 
     def add_node_from_manager(self, name: str, manager: AMDGPUManager) -> NodeInfo:
         """Discovery -> advertise -> register, in one step."""
@@ -83,6 +130,7 @@ class Cluster:
         self.core.remove_node(name)
         self.node_infos.pop(name, None)
         self.managers.pop(name, None)
+        self._drop_node_index(name)
 
     # -- scheduling --------------------------------------------------------
 
@@ -103,7 +151,7 @@ class Cluster:
         shared: Optional[PodInfo] = None
         if pod.requests.get(GPU_TOPOLOGY_GENERATION) in (None, 1) and self.node_infos:
             shared = pod.copy()
-            first_ni = self.node_infos[utils.sorted_string_keys(self.node_infos)[0]]
+            first_ni = self.node_infos[min(self.node_infos)]
             try:
                 self.scheduler.pod_allocate(first_ni, shared)
             except SchedulingError:
@@ -111,24 +159,19 @@ class Cluster:
         # Equivalence-class dedup: nodes with identical topology
         # fingerprint AND identical free-position sets produce identical
         # bind results and scores, so only one representative per class
-        # needs a (relatively expensive) trial bind.  Representative =
-        # the class member the full sort below would have picked among
-        # its (tied) members: min name for the first-fit "naive" policy,
-        # max name for the reverse (score, name) sort of "xgmi".
-        ordered = utils.sorted_string_keys(self.node_infos)
-        reps: Dict[Tuple, str] = {}
-        for name in ordered:
-            sig = self.core.state_signature(name)
-            if sig is None:
-                sig = ("__unregistered__", name)
-            if self.policy == "naive":
-                reps.setdefault(sig, name)
-            else:
-                reps[sig] = name
-        rep_names = set(reps.values())
-        for name in ordered:
-            if name not in rep_names:
-                continue
+        # needs a (relatively expensive) trial bind.  The classes are
+        # maintained INCREMENTALLY at bind/release/add/remove events, so
+        # this loop is O(distinct states), never O(nodes).
+        # Representative = the class member the full sort below would
+        # have picked among its (tied) members: min name for the
+        # first-fit "naive" policy, max name for the reverse
+        # (score, name) sort of "xgmi".
+        last = self.policy != "naive"  # naive: min name; xgmi: max name
+        rep_names = sorted(
+            members[-1] if last else members[0]
+            for members in self._classes.values()
+        )
+        for name in rep_names:
             ni = self.node_infos[name]
             # fit == "a translation + binding exists": the bind attempt
             # below subsumes the pod_fits_device predicate (which stays
@@ -166,6 +209,7 @@ class Cluster:
         pod.init_containers = bound_pod.init_containers
         pod.node_name = node_name
         self.core.take_pod_resources(node_name, pod)
+        self.reindex_node(node_name)  # its free set changed
         latency = time.perf_counter() - t0
         ring = best_score[0] if best_score else 0.0
         EVENTS.record(
@@ -186,6 +230,7 @@ class Cluster:
     def release(self, pod: PodInfo) -> None:
         if pod.node_name:
             self.core.return_pod_resources(pod.node_name, pod)
+            self.reindex_node(pod.node_name)
             # node side: clear the device plugin's in_use flags (the
             # other half of allocate's in_use=True — no write-only state)
             mgr = self.managers.get(pod.node_name)

@@ -126,6 +126,7 @@ class GroupScheduler:
         # fingerprints share one small token so per-pod signature checks
         # never re-hash the full matrix (state_signature hot path)
         self._sig_intern: Dict[Tuple, int] = {}
+        self._scorer_cache: Dict[int, TopologyScorer] = {}
         # bind-plan cache: binding is a pure function of (topology token,
         # free-position set, demand shape) at the *index* level — uuids
         # and names only enter when a plan is applied to a node.  Value:
@@ -173,7 +174,6 @@ class GroupScheduler:
             state.bw = gpus_info.bandwidth_matrix()
         else:
             state.bw = _synthetic_bw(list(state.gpus.values()))
-        state.scorer = TopologyScorer(list(state.index_to_uuid.keys()), state.bw)
         shape = tuple(
             tuple(tuple(sorted(state.gpus[u].index for u in ids)) for _, ids in g_items)
             for _, g_items in layout.groups
@@ -197,6 +197,15 @@ class GroupScheduler:
             state.topo_token = self._sig_intern.setdefault(
                 topo_sig, len(self._sig_intern)
             )
+            # One scorer per interned topology, shared by every node of
+            # that shape: subset-choice and ring memos then hit across
+            # the whole fleet (identical nodes ask identical questions),
+            # which is what keeps multi-thousand-node p50 flat.
+            scorer = self._scorer_cache.get(state.topo_token)
+            if scorer is None:
+                scorer = TopologyScorer(list(state.index_to_uuid.keys()), state.bw)
+                self._scorer_cache[state.topo_token] = scorer
+            state.scorer = scorer
             # node re-registration (watch update / re-discovery) must not
             # forget live allocations: carry over used flags for GPUs
             # that still exist (cf. the manager's in_use-survives-

@@ -411,7 +411,12 @@ class TopologyScorer:
         hit = self._ring_memo.get(key)
         if hit is not None:
             return hit
-        val, _ = best_ring(sorted(subset), self._bw)
+        if self._native is not None and all(g in self.pos for g in key):
+            val, _ = self._native.best_ring(
+                self.n, [self.pos[g] for g in key], self.flat
+            )
+        else:
+            val, _ = best_ring(sorted(subset), self._bw)
         val = min(val, 1e9)
         self._ring_memo[key] = val
         return val

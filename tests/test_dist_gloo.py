@@ -15,6 +15,16 @@ import pytest
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
+def _free_port() -> int:
+    """A fresh ephemeral port per launch: a fixed port flakes with
+    EADDRINUSE when a previous torchrun's TCPStore lingers in TIME_WAIT."""
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 def _run_dist(script: str, nproc: int = 2, timeout: int = 240, extra_env=None):
     env = dict(os.environ)
     env["MASTER_ADDR"] = "127.0.0.1"
@@ -24,7 +34,7 @@ def _run_dist(script: str, nproc: int = 2, timeout: int = 240, extra_env=None):
         [
             sys.executable, "-m", "torch.distributed.run",
             "--nnodes=1", f"--nproc-per-node={nproc}",
-            "--master-addr", "127.0.0.1", "--master-port", "29591",
+            "--master-addr", "127.0.0.1", "--master-port", str(_free_port()),
             script, *(["--gpus", str(nproc), "--steps", "3", "--warmup", "1",
                        "--pods", "50"] if script.endswith("bench.py") else []),
         ],
