@@ -171,7 +171,7 @@ void copy(at::Tensor dst, at::Tensor src) {
 // (bytes read + bytes written over wall time, hipEvent-timed).
 // blocks=0 picks the default; nontemporal selects the NT variant.
 double copy_bw_gbps(int64_t nbytes, int64_t iters, int64_t blocks_arg,
-                    bool nontemporal, int64_t variant) {
+                    bool nontemporal, int64_t variant, int64_t threads_arg) {
   TORCH_CHECK(nbytes > 0 && nbytes % 16 == 0, "nbytes must be positive, 16-aligned");
   auto opts = at::TensorOptions().dtype(at::kByte).device(at::kCUDA);
   at::Tensor src = at::empty({nbytes}, opts);
@@ -179,26 +179,30 @@ double copy_bw_gbps(int64_t nbytes, int64_t iters, int64_t blocks_arg,
   src.fill_(1);
   auto stream = at::hip::getCurrentHIPStream();
   size_t n4 = (size_t)nbytes / 16;
+  // workgroup SIZE is a runtime launch dim (the kernels read blockDim),
+  // so the wave-granularity axis can be swept without new kernels
+  int threads = threads_arg > 0 ? (int)threads_arg : BLOCK;
+  TORCH_CHECK(threads % WAVE == 0 && threads <= 1024, "threads: multiple of 64, <=1024");
   int blocks = blocks_arg > 0
                    ? (int)blocks_arg
-                   : (int)std::min<size_t>((n4 + BLOCK - 1) / BLOCK, DEFAULT_COPY_BLOCKS);
+                   : (int)std::min<size_t>((n4 + threads - 1) / threads, DEFAULT_COPY_BLOCKS);
   auto launch = [&]() {
     if (variant == 1)
-      hipLaunchKernelGGL(copy_kernel_v4_nt_u4, dim3(blocks), dim3(BLOCK), 0, stream,
+      hipLaunchKernelGGL(copy_kernel_v4_nt_u4, dim3(blocks), dim3(threads), 0, stream,
                          (const uint4v*)src.data_ptr(), (uint4v*)dst.data_ptr(), n4);
     else if (variant == 2)
-      hipLaunchKernelGGL(copy_kernel_v4_nt_chunk, dim3(blocks), dim3(BLOCK), 0,
+      hipLaunchKernelGGL(copy_kernel_v4_nt_chunk, dim3(blocks), dim3(threads), 0,
                          stream, (const uint4v*)src.data_ptr(),
                          (uint4v*)dst.data_ptr(), n4);
     else if (variant == 3)
-      hipLaunchKernelGGL(copy_kernel_v4_mixed, dim3(blocks), dim3(BLOCK), 0,
+      hipLaunchKernelGGL(copy_kernel_v4_mixed, dim3(blocks), dim3(threads), 0,
                          stream, (const uint4v*)src.data_ptr(),
                          (uint4v*)dst.data_ptr(), n4);
     else if (nontemporal)
-      hipLaunchKernelGGL(copy_kernel_v4_nt, dim3(blocks), dim3(BLOCK), 0, stream,
+      hipLaunchKernelGGL(copy_kernel_v4_nt, dim3(blocks), dim3(threads), 0, stream,
                          (const uint4v*)src.data_ptr(), (uint4v*)dst.data_ptr(), n4);
     else
-      hipLaunchKernelGGL(copy_kernel_v4, dim3(blocks), dim3(BLOCK), 0, stream,
+      hipLaunchKernelGGL(copy_kernel_v4, dim3(blocks), dim3(threads), 0, stream,
                          (const uint4*)src.data_ptr(), (uint4*)dst.data_ptr(), n4);
   };
   for (int w = 0; w < 3; ++w) launch();
@@ -297,7 +301,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("copy_bw_gbps", &copy_bw_gbps, "timed d2d copy bandwidth",
         py::call_guard<py::gil_scoped_release>(),
         py::arg("nbytes"), py::arg("iters") = 20, py::arg("blocks") = 0,
-        py::arg("nontemporal") = true, py::arg("variant") = 0);
+        py::arg("nontemporal") = true, py::arg("variant") = 0,
+        py::arg("threads") = 0);
   m.def("read_bw_gbps", &read_bw_gbps, "timed read bandwidth",
         py::call_guard<py::gil_scoped_release>(),
         py::arg("nbytes"), py::arg("iters") = 20, py::arg("blocks") = 0);
