@@ -218,3 +218,52 @@ def test_init_demand_invariants(stream):
         cluster.release(pod)
     st_node = cluster.core.nodes[node]
     assert len(st_node.free_uuids()) == total  # everything restored
+
+
+@settings(max_examples=25, deadline=None)
+@given(
+    nodes=st.lists(st.sampled_from(sorted(FIXTURES)), min_size=1, max_size=3),
+    stream=st.lists(
+        st.tuples(st.integers(min_value=1, max_value=8), st.booleans()),
+        min_size=1,
+        max_size=20,
+    ),
+)
+def test_incremental_class_index_stays_fresh(nodes, stream):
+    """The incrementally-maintained signature classes (round-2
+    O(distinct-states) scheduling) must equal freshly-computed
+    signatures after every schedule/release — stale classes would make
+    representative choice diverge from the exhaustive loop."""
+    cluster = _mk_cluster(nodes)
+
+    def check():
+        seen = set()
+        for name in cluster.node_infos:
+            fresh = cluster.core.state_signature(name)
+            if fresh is None:
+                fresh = ("__unregistered__", name)
+            assert cluster._node_sig[name] == fresh, name
+            assert name in cluster._classes[fresh]
+            seen.add(name)
+        total = sum(len(m) for m in cluster._classes.values())
+        assert total == len(seen)  # no ghost members
+
+    check()
+    live = []
+    for i, (k, rel) in enumerate(stream):
+        pod = PodInfo(
+            name=f"p{i}",
+            running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: k})},
+        )
+        try:
+            cluster.schedule(pod)
+            live.append(pod)
+        except SchedulingError:
+            pass
+        check()
+        if rel and live:
+            cluster.release(live.pop(0))
+            check()
+    for pod in live:
+        cluster.release(pod)
+    check()
