@@ -283,13 +283,13 @@ def test_choose_64_of_cpx_node_under_10ms():
     elapsed_ms = (time.perf_counter() - t0) * 1e3
     assert len(picked) == 8
     assert len({g // 8 for g in picked}) == 1  # one whole OAM
-    try:
-        import kubegpu_amd._schedcore  # noqa: F401
+    from kubegpu_amd.scheduler.xgmi import _native_available
 
-        # generous CI margin over the measured ~7 ms; the 10 ms claim is
-        # pinned by profiles/choose64_timing.json from a quiet box
+    if _native_available():
+        # generous CI margin over the measured ~2 ms on hardware; the
+        # 10 ms claim is pinned by profiles/choose64_timing_mi355x.json
         assert elapsed_ms < 200, elapsed_ms
-    except ImportError:
+    else:
         assert elapsed_ms < 5000, elapsed_ms
 
 
