@@ -47,8 +47,10 @@ class Cluster:
         # One pod binds at a time (the reference's core serializes
         # scheduler-plugin calls; without this the trial
         # bind and the commit race between threads and two pods can
-        # claim the same GPUs).
-        self._sched_lock = threading.Lock()
+        # claim the same GPUs).  RLock: the class index is mutated by
+        # add/remove/release/reindex under the same lock schedule()
+        # holds while iterating it.
+        self._sched_lock = threading.RLock()
         self.scheduler = scheduler or AMDGPUScheduler()
         self.core = GroupScheduler(policy=policy)
         self.policy = policy
@@ -77,13 +79,14 @@ class Cluster:
         gpus_info: Optional[GpusInfo] = None,
         manager: Optional[AMDGPUManager] = None,
     ) -> None:
-        name = node_info.name
-        self.scheduler.add_node(name, node_info, gpus_info)
-        self.core.register_node(node_info, gpus_info)
-        self.node_infos[name] = node_info
-        if manager is not None:
-            self.managers[name] = manager
-        self.reindex_node(name)
+        with self._sched_lock:
+            name = node_info.name
+            self.scheduler.add_node(name, node_info, gpus_info)
+            self.core.register_node(node_info, gpus_info)
+            self.node_infos[name] = node_info
+            if manager is not None:
+                self.managers[name] = manager
+            self.reindex_node(name)
 
     def reindex_node(self, name: str) -> None:
         """Re-home *name* in the signature classes after a state change.
@@ -91,6 +94,10 @@ class Cluster:
         Called automatically on add/remove/bind/release; call it
         manually only after mutating a node's core state out-of-band.
         """
+        with self._sched_lock:
+            self._reindex_locked(name)
+
+    def _reindex_locked(self, name: str) -> None:
         new_sig = self.core.state_signature(name)
         if new_sig is None:
             new_sig = ("__unregistered__", name)
@@ -126,11 +133,12 @@ class Cluster:
         return ni
 
     def remove_node(self, name: str) -> None:
-        self.scheduler.remove_node(name)
-        self.core.remove_node(name)
-        self.node_infos.pop(name, None)
-        self.managers.pop(name, None)
-        self._drop_node_index(name)
+        with self._sched_lock:
+            self.scheduler.remove_node(name)
+            self.core.remove_node(name)
+            self.node_infos.pop(name, None)
+            self.managers.pop(name, None)
+            self._drop_node_index(name)
 
     # -- scheduling --------------------------------------------------------
 
@@ -229,8 +237,9 @@ class Cluster:
 
     def release(self, pod: PodInfo) -> None:
         if pod.node_name:
-            self.core.return_pod_resources(pod.node_name, pod)
-            self.reindex_node(pod.node_name)
+            with self._sched_lock:
+                self.core.return_pod_resources(pod.node_name, pod)
+                self._reindex_locked(pod.node_name)
             # node side: clear the device plugin's in_use flags (the
             # other half of allocate's in_use=True — no write-only state)
             mgr = self.managers.get(pod.node_name)
