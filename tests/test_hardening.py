@@ -363,3 +363,58 @@ def test_json_log_mode(tmp_path):
     lines = [json.loads(l) for l in out.stderr.strip().splitlines()]
     assert lines[0]["msg"] == "hello world" and lines[0]["level"] == "info"
     assert lines[1]["msg"] == "bad 7" and lines[1]["level"] == "error"
+
+
+def test_concurrent_schedule_with_node_churn():
+    """Schedules racing node add/remove (watch events in production):
+    the incremental class index must stay consistent and nothing may
+    corrupt — the round-2 index is mutated under the scheduling lock."""
+    cluster = Cluster()
+    for n in range(4):
+        mgr = create_device_plugin(FakeBackend(fixtures.fixture_8x_mi355x()))
+        cluster.add_node_from_manager(f"stable{n}", mgr)
+    errors = []
+    stop = threading.Event()
+
+    def churner():
+        try:
+            i = 0
+            while not stop.is_set():
+                mgr = create_device_plugin(FakeBackend(fixtures.fixture_8x_mi355x()))
+                cluster.add_node_from_manager(f"churn{i % 3}", mgr)
+                cluster.remove_node(f"churn{i % 3}")
+                i += 1
+        except Exception as e:  # pragma: no cover
+            errors.append(e)
+
+    def scheduler_worker(tid):
+        try:
+            for i in range(40):
+                pod = PodInfo(
+                    name=f"c{tid}-{i}",
+                    running_containers={
+                        "c": ContainerInfo(kube_requests={RESOURCE_GPU: 2})
+                    },
+                )
+                try:
+                    cluster.schedule(pod)
+                except SchedulingError:
+                    continue
+                cluster.release(pod)
+        except Exception as e:  # pragma: no cover
+            errors.append(e)
+
+    ch = threading.Thread(target=churner)
+    workers = [threading.Thread(target=scheduler_worker, args=(t,)) for t in range(4)]
+    ch.start()
+    for t in workers:
+        t.start()
+    for t in workers:
+        t.join()
+    stop.set()
+    ch.join()
+    assert not errors
+    # index consistent with live nodes
+    assert set(cluster._node_sig) == set(cluster.node_infos)
+    for n in range(4):
+        assert cluster.core.free_count(f"stable{n}") == 8
