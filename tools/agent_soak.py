@@ -86,7 +86,13 @@ def main() -> int:
     )
     ids = sorted(mgr.gpus)
     calls = allocs = errors = 0
-    lat = []
+    # bounded reservoir: client-side sample storage must not read as a
+    # server "leak" in the RSS numbers (both live in this process)
+    import random
+
+    reservoir: list = []
+    RESERVOIR = 20000
+    rng = random.Random(0)
     while time.monotonic() < stop:
         k = [1, 2, 4, min(8, len(ids))][calls % 4]
         t0 = time.perf_counter()
@@ -108,11 +114,17 @@ def main() -> int:
             allocs += 1
         except Exception:
             errors += 1
-        lat.append(time.perf_counter() - t0)
+        dt = time.perf_counter() - t0
+        if len(reservoir) < RESERVOIR:
+            reservoir.append(dt)
+        else:
+            j = rng.randrange(calls + 1)
+            if j < RESERVOIR:
+                reservoir[j] = dt
         calls += 1
     ch.close()
     plugin.stop()
-    lat.sort()
+    lat = sorted(reservoir)
     out = {
         "duration_s": args.seconds,
         "backend": "fake" if args.fake else "real",
@@ -120,6 +132,7 @@ def main() -> int:
         "preferred_plus_allocate_pairs": calls,
         "alloc_ok": allocs,
         "errors": errors,
+        "latency_sample": len(lat),
         "rpc_pair_p50_ms": round(lat[len(lat) // 2] * 1e3, 3) if lat else None,
         "rpc_pair_p99_ms": round(lat[int(len(lat) * 0.99)] * 1e3, 3) if lat else None,
         "rss_start_mb": round(rss0, 1),
