@@ -181,17 +181,24 @@ def translate_pod_gpu_resources(
 ) -> None:
     """Dispatch on the gpu-generate-topology knob (gpu.go:94-127).
 
-    Raises SchedulingError when no translation exists.
+    Exact reference semantics (gpu.go:104-123): knob unset/1 tries the
+    best cached tree, and when NO tree holds the demand it falls back
+    to the flat/no-topology translation instead of failing — whether
+    the pod then fits is the group core's per-node decision.  knob 0
+    goes flat directly; any other value is an error.
     """
     knob = pod.requests.get(GPU_TOPOLOGY_GENERATION)
     if knob is None or knob == 1:
-        convert_to_best_gpu_requests(pod, cache)
-    elif knob == 0:
-        for cont in list(pod.running_containers.values()) + list(
-            pod.init_containers.values()
-        ):
-            translate_gpu_container_resources(node_info.allocatable, cont)
-    else:
+        try:
+            convert_to_best_gpu_requests(pod, cache)
+            return
+        except SchedulingError:
+            pass  # !found -> flat fallback (gpu.go:113-116)
+    elif knob != 0:
         raise SchedulingError(
             f"invalid {GPU_TOPOLOGY_GENERATION} value {knob} for pod {pod.name}"
         )
+    for cont in list(pod.running_containers.values()) + list(
+        pod.init_containers.values()
+    ):
+        translate_gpu_container_resources(node_info.allocatable, cont)

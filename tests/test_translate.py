@@ -99,16 +99,40 @@ def test_topology_knob_dispatch():
         translate_pod_gpu_resources(ni, podx, cache)
 
 
-def test_no_tree_big_pod_fails():
+def test_no_tree_big_pod_falls_back_to_flat():
+    """When no cached tree holds the demand, translation falls back to
+    the flat/no-topology path instead of erroring — reference semantics
+    (gpu.go:104-116: `if !found || req == 0` both land on the flat
+    loop).  Whether the pod then fits is the group core's decision."""
     cache = NodeTreeCache()
-    _cached_node(cache)
+    _cached_node(cache)  # 8-card tree, demand is 9
     ni = NodeInfo(name="node0")
     pod = PodInfo(
         name="big",
         running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: 9})},
     )
+    translate_pod_gpu_resources(ni, pod, cache)
+    reqs = pod.running_containers["c"].dev_requests
+    assert len(reqs) == 9
+    assert all("/gpu/" in r and r.endswith("/cards") for r in reqs)
+
+
+def test_big_pod_still_unschedulable_end_to_end():
+    """The flat fallback does not make an oversized pod schedulable: the
+    binder has only 8 concrete GPUs."""
+    from kubegpu_amd.core import Cluster
+    from kubegpu_amd.deviceplugin import create_device_plugin
+    from kubegpu_amd.discovery import FakeBackend, fixtures
+
+    cluster = Cluster()
+    mgr = create_device_plugin(FakeBackend(fixtures.fixture_8x_mi355x()))
+    cluster.add_node_from_manager("n0", mgr)
+    pod = PodInfo(
+        name="big",
+        running_containers={"c": ContainerInfo(kube_requests={RESOURCE_GPU: 9})},
+    )
     with pytest.raises(SchedulingError):
-        translate_pod_gpu_resources(ni, pod, cache)
+        cluster.schedule(pod)
 
 
 def test_multi_container_disjoint_slots():
