@@ -23,4 +23,4 @@ probe           RCCL-over-xGMI bandwidth probe + HIP HBM bandwidth kernels
 cli             operator CLIs (cf. nvidiagpuplugin/cmd/, nvmlinfo/)
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
