@@ -65,6 +65,7 @@ class Cluster:
 
         self._sorted_set = SortedSet
         self._node_sig: Dict[str, Tuple] = {}
+        self._first_node: Optional[str] = None  # cached min(node_infos)
         # SortedSet members: O(log n) add/discard, O(1) min/max, so the
         # per-class representative (the member the full candidate sort
         # would pick) costs nothing to maintain even when scheduling
@@ -86,6 +87,8 @@ class Cluster:
             self.node_infos[name] = node_info
             if manager is not None:
                 self.managers[name] = manager
+            if self._first_node is None or name < self._first_node:
+                self._first_node = name
             self.reindex_node(name)
 
     def reindex_node(self, name: str) -> None:
@@ -138,6 +141,8 @@ class Cluster:
             self.core.remove_node(name)
             self.node_infos.pop(name, None)
             self.managers.pop(name, None)
+            if name == self._first_node:
+                self._first_node = min(self.node_infos) if self.node_infos else None
             self._drop_node_index(name)
 
     # -- scheduling --------------------------------------------------------
@@ -159,7 +164,7 @@ class Cluster:
         shared: Optional[PodInfo] = None
         if pod.requests.get(GPU_TOPOLOGY_GENERATION) in (None, 1) and self.node_infos:
             shared = pod.copy()
-            first_ni = self.node_infos[min(self.node_infos)]
+            first_ni = self.node_infos[self._first_node]
             try:
                 self.scheduler.pod_allocate(first_ni, shared)
             except SchedulingError:
