@@ -61,3 +61,27 @@ def test_real_payload_through_full_stack():
     mounts, devices, envs = cluster.container_allocate(pod, "c")
     assert devices == ["/dev/kfd", "/dev/dri/renderD184", "/dev/dri/card56"]
     assert envs["ROCR_VISIBLE_DEVICES"] == "93ff75a3-0000-1000-8091-62812952020a"
+
+
+def test_round2_capture_parses_with_runtime_versions():
+    """Round-2 verbatim capture (`amdsmiinfo json`, gpurun session F):
+    the version block is runtime-queried — rocm from
+    /opt/rocm/.info/version, amdsmi via amdsmi_get_lib_version — not the
+    round-1 hardcode (VERDICT #7)."""
+    path = os.path.join(os.path.dirname(__file__), "fixtures",
+                        "real_1x_mi355x_r2.json")
+    with open(path) as f:
+        raw = json.load(f)
+    assert raw["version"]["rocm"] == "7.2.0"
+    assert raw["version"]["amdsmi"] == "26.2.1"  # a real lib version
+    info = GpusInfo.from_json(json.dumps(raw))
+    g = info.devices[0]
+    assert g.gfx_target == "gfx950"
+    assert g.compute_units == 256
+    assert g.memory.vram_total_bytes > 300e9
+    # drives the manager end to end like the round-1 capture
+    mgr = create_device_plugin(FakeBackend(info))
+    mgr.start()
+    ni = NodeInfo(name="captured-r2")
+    mgr.update_node_info(ni)
+    assert ni.kube_alloc[RESOURCE_GPU] == 1
