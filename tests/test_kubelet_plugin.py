@@ -712,3 +712,47 @@ def test_plugin_socket_recreation_race(tmp_path):
     finally:
         p2.stop()
         p1.stop()
+
+
+@pytest.mark.timeout(120)
+def test_agent_process_lifecycle_with_fake_kubelet(tmp_path):
+    """The production entry point end to end as a real process: starts,
+    registers with a (fake) kubelet, survives a kubelet restart via the
+    watcher, and exits 0 on SIGTERM."""
+    import signal
+    import subprocess
+    import sys
+    import time as _time
+
+    kubelet = _FakeKubelet(str(tmp_path / "kubelet.sock"))
+    kubelet.start()
+    try:
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "kubegpu_amd.server.agent",
+             "--fake",
+             "--socket", str(tmp_path / "agent.sock"),
+             "--kubelet-socket", kubelet.sock_path,
+             "--metrics-port", "0",
+             "--health-interval", "0.2"],
+            cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+        )
+        assert kubelet.event.wait(30), "agent never registered"
+
+        # kubelet restart: agent's watcher must re-register
+        before = len(kubelet.registrations)
+        kubelet.stop()
+        _time.sleep(0.5)
+        kubelet.start()
+        deadline = _time.time() + 20
+        while len(kubelet.registrations) <= before and _time.time() < deadline:
+            _time.sleep(0.1)
+        assert len(kubelet.registrations) > before, "no re-registration"
+
+        proc.send_signal(signal.SIGTERM)
+        out, _ = proc.communicate(timeout=30)
+        assert proc.returncode == 0, out[-1500:]
+    finally:
+        kubelet.stop()
+        if proc.poll() is None:
+            proc.kill()
