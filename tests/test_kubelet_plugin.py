@@ -756,3 +756,38 @@ def test_agent_process_lifecycle_with_fake_kubelet(tmp_path):
         kubelet.stop()
         if proc.poll() is None:
             proc.kill()
+
+
+@pytest.mark.timeout(120)
+def test_watcher_retries_until_kubelet_accepts(tmp_path):
+    """A kubelet that comes back up broken (rejecting registrations)
+    then recovers: the watcher keeps retrying each tick until accepted
+    instead of giving up after the first failure."""
+    import time as _time
+
+    kubelet = _FakeKubelet(str(tmp_path / "kubelet.sock"))
+    kubelet.start()
+    mgr = create_device_plugin(FakeBackend(fixtures.fixture_8x_mi355x()))
+    mgr.start()
+    p = KubeletDevicePlugin(mgr, socket_path=str(tmp_path / "w.sock"))
+    p.start()
+    try:
+        p.register_with_kubelet(kubelet.sock_path)
+        p.watch_kubelet(kubelet.sock_path, interval_s=0.1)
+        before = len(kubelet.registrations)
+
+        # restart REJECTING v1beta1 (config error), then fix it
+        kubelet.stop()
+        _time.sleep(0.3)
+        kubelet.accept_versions = ("v1alpha2",)
+        kubelet.start()
+        _time.sleep(0.6)  # several failed watcher attempts
+        assert len(kubelet.registrations) == before  # rejected, none landed
+        kubelet.accept_versions = ("v1beta1",)  # kubelet fixed (no restart)
+        deadline = _time.time() + 10
+        while len(kubelet.registrations) <= before and _time.time() < deadline:
+            _time.sleep(0.05)
+        assert len(kubelet.registrations) > before, "watcher gave up"
+    finally:
+        p.stop()
+        kubelet.stop()
