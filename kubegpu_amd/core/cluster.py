@@ -19,7 +19,6 @@ import time
 from dataclasses import dataclass
 from typing import Dict, List, Optional, Tuple
 
-from ..api import utils
 from ..api.types import NodeInfo, PodInfo
 from ..deviceplugin.manager import AMDGPUManager
 from ..discovery import GpusInfo
