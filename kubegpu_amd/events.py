@@ -37,6 +37,9 @@ class EventTrace:
         self._pending: Deque[str] = deque(maxlen=8192)
         self._flush_wake = threading.Event()
         self._flusher: Optional[threading.Thread] = None
+        # serializes grab+write so concurrent drains (flusher thread +
+        # an explicit flush()) cannot append batches out of order
+        self._drain_lock = threading.Lock()
 
     def record(self, event: str, **fields) -> None:
         rec = {"ts": round(time.time(), 3), "event": event, **fields}
@@ -65,17 +68,18 @@ class EventTrace:
             self._drain()
 
     def _drain(self) -> None:
-        with self._lock:
-            if not self._pending or not self._path:
-                return
-            lines = list(self._pending)
-            self._pending.clear()
-            path = self._path
-        try:
-            with open(path, "a") as f:
-                f.write("\n".join(lines) + "\n")
-        except OSError:
-            pass  # best-effort: never affect scheduling
+        with self._drain_lock:
+            with self._lock:
+                if not self._pending or not self._path:
+                    return
+                lines = list(self._pending)
+                self._pending.clear()
+                path = self._path
+            try:
+                with open(path, "a") as f:
+                    f.write("\n".join(lines) + "\n")
+            except OSError:
+                pass  # best-effort: never affect scheduling
 
     def flush(self, timeout_s: float = 5.0) -> None:
         """Synchronously drain pending file writes (tests/shutdown)."""

@@ -418,3 +418,21 @@ def test_concurrent_schedule_with_node_churn():
     assert set(cluster._node_sig) == set(cluster.node_infos)
     for n in range(4):
         assert cluster.core.free_count(f"stable{n}") == 8
+
+
+def test_event_flusher_under_load(tmp_path):
+    """The async event-file flusher loses nothing under a
+    burst well past its wake batching, and every line is valid JSON."""
+    from kubegpu_amd import events
+
+    log = tmp_path / "burst.jsonl"
+    trace = events.EventTrace(capacity=64, path=str(log))
+    for i in range(5000):
+        trace.record("schedule", pod=f"p{i}", node="n0", gpus=["g"], latency_ms=0.1)
+    trace.flush()
+    lines = log.read_text().splitlines()
+    assert len(lines) == 5000
+    recs = [json.loads(l) for l in lines[-10:]]
+    assert recs[-1]["pod"] == "p4999"
+    # ring stays bounded regardless
+    assert len(trace.recent(1000)) == 64
