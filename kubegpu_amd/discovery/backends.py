@@ -225,7 +225,15 @@ class SysfsBackend(Backend):
                     )
                 )
             gpus.append(g)
-        info = GpusInfo(version=VersionInfo(rocm="sysfs"), devices=gpus)
+        rocm = "sysfs"
+        try:
+            with open("/opt/rocm/.info/version") as f:
+                v = f.read().strip()
+                if v:
+                    rocm = v
+        except OSError:
+            pass
+        info = GpusInfo(version=VersionInfo(rocm=rocm), devices=gpus)
         return info.to_json().encode()
 
 
