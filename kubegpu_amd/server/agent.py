@@ -32,6 +32,10 @@ def main(argv=None) -> int:
     ap.add_argument("--socket", default=None)
     ap.add_argument("--plugin-dir", default=dpapi.DEVICE_PLUGIN_PATH)
     ap.add_argument("--kubelet-socket", default=dpapi.KUBELET_SOCKET)
+    ap.add_argument("--pod-resources-socket", default=None,
+                    help="kubelet pod-resources socket for in_use "
+                         "reconciliation (default: the standard path "
+                         "when it exists)")
     ap.add_argument("--no-register", action="store_true")
     ap.add_argument("--metrics-port", type=int, default=9400)
     ap.add_argument("--health-interval", type=float, default=30.0)
@@ -115,6 +119,16 @@ def main(argv=None) -> int:
             plugin.start()
             registered = False
         plugin.servicer.notify()  # wake ListAndWatch to refresh health
+        # kubelet is the allocation source of truth on the stock path:
+        # reconcile in_use from its pod-resources API when available
+        try:
+            from .podresources import PodResourcesClient, reconcile_in_use
+
+            client = (PodResourcesClient(args.pod_resources_socket)
+                      if args.pod_resources_socket else PodResourcesClient())
+            reconcile_in_use(manager, client)
+        except Exception as e:  # never let reconcile kill the agent
+            utils.logf(2, "agent: pod-resources reconcile error: %s", e)
         for uuid, ok in manager.device_health().items():
             g = manager.gpu_or_tombstone(uuid)
             METRICS.set_gpu_health(
