@@ -75,6 +75,10 @@ def _build():
     msg("ListPodResourcesResponse", [
         ("pod_resources", 1, M, REP, ref("PodResources")),
     ])
+    msg("AllocatableResourcesRequest", [])
+    msg("AllocatableResourcesResponse", [
+        ("devices", 1, M, REP, ref("ContainerDevices")),
+    ])
 
     pool = descriptor_pool.DescriptorPool()
     pool.Add(fdp)
@@ -90,6 +94,8 @@ _MESSAGES = _build()
 
 ListPodResourcesRequest = _MESSAGES["ListPodResourcesRequest"]
 ListPodResourcesResponse = _MESSAGES["ListPodResourcesResponse"]
+AllocatableResourcesRequest = _MESSAGES["AllocatableResourcesRequest"]
+AllocatableResourcesResponse = _MESSAGES["AllocatableResourcesResponse"]
 ContainerDevices = _MESSAGES["ContainerDevices"]
 ContainerResources = _MESSAGES["ContainerResources"]
 PodResources = _MESSAGES["PodResources"]
@@ -115,6 +121,21 @@ class PodResourcesClient:
                 response_deserializer=ListPodResourcesResponse.FromString,
             )
             return rpc(ListPodResourcesRequest(), timeout=timeout_s)
+        finally:
+            channel.close()
+
+    def get_allocatable_resources(self, timeout_s: float = 10.0):
+        """kubelet's view of allocatable devices (v1
+        GetAllocatableResources) — lets a monitor cross-check that the
+        devices we advertise are the ones kubelet accounts."""
+        channel = grpc.insecure_channel(f"unix://{self.socket_path}")
+        try:
+            rpc = channel.unary_unary(
+                f"/{POD_RESOURCES_SERVICE}/GetAllocatableResources",
+                request_serializer=lambda m: m.SerializeToString(),
+                response_deserializer=AllocatableResourcesResponse.FromString,
+            )
+            return rpc(AllocatableResourcesRequest(), timeout=timeout_s)
         finally:
             channel.close()
 

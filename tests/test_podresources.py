@@ -160,3 +160,31 @@ def test_idle_preference_follows_reconciled_state(kubelet_pr, tmp_path):
         ch.close()
     finally:
         p.stop()
+
+
+def test_get_allocatable_resources_roundtrip(kubelet_pr):
+    """GetAllocatableResources client against a fake kubelet endpoint."""
+    # extend the fake with the second RPC
+    srv = kubelet_pr
+
+    def alloc_handler(request, context):
+        resp = pr.AllocatableResourcesResponse()
+        d = resp.devices.add()
+        d.resource_name = "amd.com/gpu"
+        d.device_ids.extend([f"GPU-mi355x-{i:02d}" for i in range(8)])
+        return resp
+
+    srv.server.add_generic_rpc_handlers((
+        grpc.method_handlers_generic_handler(
+            pr.POD_RESOURCES_SERVICE,
+            {"GetAllocatableResources": grpc.unary_unary_rpc_method_handler(
+                alloc_handler,
+                request_deserializer=pr.AllocatableResourcesRequest.FromString,
+                response_serializer=lambda m: m.SerializeToString(),
+            )},
+        ),
+    ))
+    client = pr.PodResourcesClient(srv.sock_path)
+    resp = client.get_allocatable_resources()
+    assert len(resp.devices) == 1
+    assert len(resp.devices[0].device_ids) == 8
