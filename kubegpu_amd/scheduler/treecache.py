@@ -88,27 +88,30 @@ def parse_node_resources(resources: Dict[str, int]) -> Tuple[Optional[SortedTree
     return root, layout
 
 
-def _height(node: SortedTreeNode) -> int:
-    if not node.children:
-        return 1
-    return 1 + max(_height(c) for c in node.children)
-
-
 def compute_tree_score(node: SortedTreeNode) -> float:
     """Score a tree: denser, deeper grouping scores higher.
 
-    Reference formula (gpu.go:180-190): score at a level =
-    val * level / numChildren, summed recursively — a node whose GPUs sit
-    in fewer, larger groups (better interconnect locality) outranks a
-    fragmented one of equal size.
+    EXACT reference recursion (computeTreeScoreAtLevel, gpu.go:180-190):
+
+        score(n, level, numChild_of_parent) =
+            n.val * level / numChild_of_parent
+            + Σ score(child, level + 1, len(n.children))
+
+    entered at the root with level=0 and numChild=len(root.children)
+    (so the root's own term is 0).  Deeper levels weigh more and each
+    term is divided by its sibling count — fewer, larger groups
+    (better interconnect locality) outrank a fragmented shape of equal
+    size.  Each node's .score records its own subtree contribution.
     """
-    level = _height(node)
-    if not node.children:
-        node.score = float(node.val)
-        return node.score
-    s = sum(compute_tree_score(c) for c in node.children)
-    node.score = s + node.val * level / len(node.children)
-    return node.score
+
+    def at_level(n: SortedTreeNode, level: int, num_child: int) -> float:
+        s = (n.val * level) / num_child if num_child else 0.0
+        for c in n.children:
+            s += at_level(c, level + 1, len(n.children))
+        n.score = s
+        return s
+
+    return at_level(node, 0, len(node.children) or 1)
 
 
 def tree_key(node: Optional[SortedTreeNode]) -> str:

@@ -87,3 +87,27 @@ def test_node_shape_change_rehomes():
     key2 = cache.node_location_map["n"]
     assert key1 != key2
     assert len(cache) == 1  # old shape garbage-collected
+
+
+def test_score_formula_matches_reference_exactly():
+    """Pin the exact reference recursion (computeTreeScoreAtLevel,
+    gpu.go:180-190): score = Σ val·level/numChild_of_parent over the
+    tree, root entered at level 0.
+
+    Hand-computed: one 8-dense gpugrp0 → 8·1/1 + 8·2/1 = 24;
+    two 4-hives → 8·1/1 + 2·(4·2/2) = 16;
+    four 2-pairs → 8·1/1 + 4·(2·2/4) = 12."""
+    def res_for(group_sizes):
+        res, g = {}, 0
+        for gi, size in enumerate(group_sizes):
+            for _ in range(size):
+                res[f"resource/group/gpugrp1/0/gpugrp0/{gi}/gpu/G{g}/cards"] = 1
+                g += 1
+        return res
+
+    dense, _ = parse_node_resources(res_for([8]))
+    split, _ = parse_node_resources(res_for([4, 4]))
+    frag, _ = parse_node_resources(res_for([2, 2, 2, 2]))
+    assert dense.score == 24.0
+    assert split.score == 16.0
+    assert frag.score == 12.0
