@@ -32,6 +32,9 @@ def main(argv=None) -> int:
     p.add_argument("--bytes", type=int, default=256 << 20)
     p.add_argument("--health", action="store_true",
                    help="per-GPU health view (ECC totals, tombstones)")
+    p.add_argument("--cdi", action="store_true",
+                   help="emit a CDI v0.6.0 spec for the node's GPUs "
+                        "(write to /etc/cdi/amd.com-gpu.json to activate)")
     p.add_argument("--fake", action="store_true",
                    help="use the 8xMI355X fixture backend (no GPU needed)")
     from .. import __version__
@@ -46,6 +49,17 @@ def main(argv=None) -> int:
         backend = FakeBackend(fixtures.fixture_8x_mi355x())
     else:
         backend = default_backend()
+    if args.cdi:
+        from ..deviceplugin.cdi import cdi_json
+
+        info = None
+        mgr = create_device_plugin(backend)
+        mgr.start()
+        if mgr._last_info is None:
+            print("no GPUs discovered", file=sys.stderr)
+            return 1
+        print(cdi_json(mgr._last_info))
+        return 0
     if args.health:
         mgr = create_device_plugin(backend)
         mgr.start()

@@ -376,3 +376,35 @@ def test_forced_refresh_waits_for_inflight_fetch():
     assert not t2.is_alive()
     assert done["gen"] >= 2  # observed the completed refresh
     assert slow.calls == 2  # waiter reused the in-flight fetch
+
+
+def test_cdi_spec_structure():
+    """CDI v0.6.0 spec from discovery: per-GPU render/card nodes plus a
+    common /dev/kfd edit and index aliases (containerd/CRI-O consume
+    this directly — the no-hook injection path, north star)."""
+    from kubegpu_amd.deviceplugin.cdi import cdi_spec
+
+    info = fixtures.fixture_2hive_8gpu()
+    spec = cdi_spec(info)
+    assert spec["cdiVersion"] == "0.6.0"
+    assert spec["kind"] == "amd.com/gpu"
+    kfd = spec["containerEdits"]["deviceNodes"]
+    assert kfd and kfd[0]["path"] == "/dev/kfd"
+    names = {d["name"] for d in spec["devices"]}
+    assert "GPU-mi355x-00" in names and "0" in names  # uuid + index alias
+    by_name = {d["name"]: d for d in spec["devices"]}
+    edits = by_name["GPU-mi355x-03"]["containerEdits"]
+    assert any("renderD" in n["path"] for n in edits["deviceNodes"])
+    assert edits["env"] == ["ROCR_VISIBLE_DEVICES=GPU-mi355x-03"]
+    # 8 GPUs x (uuid + index alias)
+    assert len(spec["devices"]) == 16
+
+
+def test_amddevs_cdi_cli(capsys):
+    import json as _json
+
+    from kubegpu_amd.cli.amddevs import main as amddevs_main
+
+    assert amddevs_main(["--fake", "--cdi"]) == 0
+    spec = _json.loads(capsys.readouterr().out)
+    assert spec["kind"] == "amd.com/gpu" and len(spec["devices"]) == 16
