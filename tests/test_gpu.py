@@ -293,3 +293,39 @@ def test_rcclprobe_across_all_gpus_when_multi():
     # xGMI ring floor: well below one link (~153 GB/s) means the ring
     # fell back to host paths — the placement model would be wrong
     assert rec["busbw_gbps"] > 60, (rec, {"predicted": pred})
+
+
+def test_extender_with_real_inventory():
+    """Scheduler-extender webhook fed the REAL box inventory over HTTP:
+    register via POST /v1/nodes/<n> with live amdsmiinfo json, then
+    filter + prioritize a 1-GPU pod."""
+    import urllib.request
+
+    from kubegpu_amd.server.extender import serve
+
+    server, core = serve(host="127.0.0.1", port=0)
+    port = server.server_address[1]
+    base = f"http://127.0.0.1:{port}"
+
+    def post(path, payload):
+        req = urllib.request.Request(
+            base + path, data=json.dumps(payload).encode(),
+            headers={"Content-Type": "application/json"}, method="POST")
+        with urllib.request.urlopen(req, timeout=30) as r:
+            return json.loads(r.read().decode())
+
+    try:
+        out = subprocess.run([os.path.join(BIN, "amdsmiinfo"), "json"],
+                             capture_output=True, timeout=120)
+        inv = json.loads(out.stdout)
+        reg = post("/v1/nodes/realnode", inv)
+        assert reg["registered"] == "realnode" and reg["gpus"] >= 1
+
+        pod = {"metadata": {"name": "p"}, "spec": {"containers": [
+            {"name": "c", "resources": {"limits": {"amd.com/gpu": "1"}}}]}}
+        res = post("/v1/filter", {"Pod": pod, "NodeNames": ["realnode"]})
+        assert res["NodeNames"] == ["realnode"], res
+        pri = post("/v1/prioritize", {"Pod": pod, "NodeNames": ["realnode"]})
+        assert pri[0]["Score"] == 10
+    finally:
+        server.shutdown()
