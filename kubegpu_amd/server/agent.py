@@ -37,6 +37,14 @@ def main(argv=None) -> int:
                          "reconciliation (default: the standard path "
                          "when it exists)")
     ap.add_argument("--no-register", action="store_true")
+    ap.add_argument("--extender-url", default=None,
+                    help="scheduler-extender base URL (e.g. "
+                         "http://gpu-extender:9109); the agent POSTs its "
+                         "inventory to /v1/nodes/<node-name> every health "
+                         "tick so the extender scores this node")
+    ap.add_argument("--node-name", default=None,
+                    help="node name for extender registration "
+                         "(default: hostname)")
     ap.add_argument("--metrics-port", type=int, default=9400)
     ap.add_argument("--health-interval", type=float, default=30.0)
     ap.add_argument("--fake", action="store_true",
@@ -119,6 +127,21 @@ def main(argv=None) -> int:
             plugin.start()
             registered = False
         plugin.servicer.notify()  # wake ListAndWatch to refresh health
+        if args.extender_url and manager._last_info is not None:
+            try:
+                import socket as _socket
+                import urllib.request as _ur
+
+                node = args.node_name or _socket.gethostname()
+                req = _ur.Request(
+                    f"{args.extender_url.rstrip('/')}/v1/nodes/{node}",
+                    data=manager._last_info.to_json().encode(),
+                    headers={"Content-Type": "application/json"},
+                    method="POST",
+                )
+                _ur.urlopen(req, timeout=10).read()
+            except Exception as e:
+                utils.logf(2, "agent: extender registration failed: %s", e)
         # kubelet is the allocation source of truth on the stock path:
         # reconcile in_use from its pod-resources API when available
         try:

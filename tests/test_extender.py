@@ -123,3 +123,43 @@ def test_extender_http_roundtrip():
         assert res["NodeNames"] == []
     finally:
         server.shutdown()
+
+
+@pytest.mark.timeout(120)
+def test_agent_feeds_extender(tmp_path):
+    """Deployment loop end to end: the node agent POSTs its inventory to
+    the extender every health tick; the extender then scores the node
+    for kube-scheduler."""
+    import signal
+    import subprocess
+    import sys
+    import time as _time
+    import os
+
+    server, core = serve(host="127.0.0.1", port=0)
+    port = server.server_address[1]
+    try:
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "kubegpu_amd.server.agent",
+             "--fake", "--no-register",
+             "--socket", str(tmp_path / "agent.sock"),
+             "--metrics-port", "0",
+             "--health-interval", "0.2",
+             "--extender-url", f"http://127.0.0.1:{port}",
+             "--node-name", "fed-node"],
+            cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+        )
+        deadline = _time.time() + 20
+        while "fed-node" not in core.cluster.node_infos and _time.time() < deadline:
+            _time.sleep(0.1)
+        assert "fed-node" in core.cluster.node_infos, "agent never registered"
+        res = core.filter({"Pod": _pod(4), "NodeNames": ["fed-node"]})
+        assert res["NodeNames"] == ["fed-node"]
+        proc.send_signal(signal.SIGTERM)
+        out, _ = proc.communicate(timeout=30)
+        assert proc.returncode == 0, out[-1000:]
+    finally:
+        server.shutdown()
+        if proc.poll() is None:
+            proc.kill()
