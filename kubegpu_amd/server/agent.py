@@ -14,6 +14,7 @@ Usage:
 from __future__ import annotations
 
 import argparse
+import json
 import os
 import signal
 import sys
@@ -133,9 +134,11 @@ def main(argv=None) -> int:
                 import urllib.request as _ur
 
                 node = args.node_name or _socket.gethostname()
+                payload = json.loads(manager._last_info.to_json())
+                payload["in_use"] = manager.in_use_uuids()
                 req = _ur.Request(
                     f"{args.extender_url.rstrip('/')}/v1/nodes/{node}",
-                    data=manager._last_info.to_json().encode(),
+                    data=json.dumps(payload).encode(),
                     headers={"Content-Type": "application/json"},
                     method="POST",
                 )

@@ -74,7 +74,19 @@ class ExtenderCore:
 
     # -- node registry -----------------------------------------------------
 
-    def register_node(self, name: str, gpus_info: GpusInfo) -> None:
+    def register_node(
+        self, name: str, gpus_info: GpusInfo, in_use: List[str] = ()
+    ) -> None:
+        """Register/refresh a node.
+
+        Occupancy comes WITH the registration: the agent reports which
+        uuids hold live allocations (its in_use view, reconciled from
+        kubelet's pod-resources API) as a top-level ``in_use`` list,
+        and any GPU with live compute processes counts as occupied too
+        — the extender itself never observes bindings (kube-scheduler
+        does not call back after scheduling), so each refresh carries
+        the node's current truth.
+        """
         with self._lock:
             ni = NodeInfo(name=name)
             # advertise exactly like the device plugin would
@@ -87,6 +99,12 @@ class ExtenderCore:
             if name in self.cluster.node_infos:
                 self.cluster.remove_node(name)
             self.cluster.add_node(ni, mgr._last_info, mgr)
+            state = self.cluster.core.nodes[name]
+            occupied = set(in_use)
+            for g in gpus_info.devices:
+                if g.uuid in occupied or (g.process_count or 0) > 0:
+                    state.mark_used(g.uuid)
+            self.cluster.reindex_node(name)
 
     def remove_node(self, name: str) -> None:
         with self._lock:
@@ -200,10 +218,13 @@ class _Handler(BaseHTTPRequestHandler):
             else:
                 m = _NODE_RE.match(self.path)
                 if m:
-                    info = GpusInfo.from_json(json.dumps(self._body()))
-                    self.core.register_node(m.group(1), info)
+                    body = self._body()
+                    in_use = body.get("in_use") or []
+                    info = GpusInfo.from_json(json.dumps(body))
+                    self.core.register_node(m.group(1), info, in_use)
                     self._reply(200, {"registered": m.group(1),
-                                      "gpus": len(info.devices)})
+                                      "gpus": len(info.devices),
+                                      "in_use": len(in_use)})
                 else:
                     self._reply(404, {"error": "not found"})
         except Exception as e:  # malformed input must not kill the server

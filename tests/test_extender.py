@@ -94,7 +94,7 @@ def test_extender_http_roundtrip():
     try:
         fix = json.loads(fixtures.fixture_2hive_8gpu().to_json())
         out = post("/v1/nodes/n0", fix)
-        assert out == {"registered": "n0", "gpus": 8}
+        assert out == {"registered": "n0", "gpus": 8, "in_use": 0}
 
         res = post("/v1/filter", {"Pod": _pod(4), "NodeNames": ["n0", "nope"]})
         assert res["NodeNames"] == ["n0"]
@@ -163,3 +163,36 @@ def test_agent_feeds_extender(tmp_path):
         server.shutdown()
         if proc.poll() is None:
             proc.kill()
+
+
+def test_registration_carries_occupancy():
+    """The agent's in_use list (reconciled from kubelet pod-resources)
+    and live process counts flow into the extender's occupancy: an
+    8-GPU pod no longer fits a node with 2 GPUs held, and a 4-GPU pod
+    avoids the held hive; a later refresh with everything free clears
+    the state."""
+    core = ExtenderCore()
+    fix = fixtures.fixture_2hive_8gpu()
+    core.register_node("n0", fix,
+                       in_use=["GPU-mi355x-00", "GPU-mi355x-01"])
+
+    res = core.filter({"Pod": _pod(8), "NodeNames": ["n0"]})
+    assert res["NodeNames"] == []  # only 6 free
+
+    res = core.filter({"Pod": _pod(4), "NodeNames": ["n0"]})
+    assert res["NodeNames"] == ["n0"]
+    # the chosen subset for 4 must be the intact hive 1
+    bw = core._trial("p", 4, "n0")
+    assert bw and bw > 100  # xGMI-class, not PCIe-bound
+
+    # process_count also counts as occupied
+    fix2 = fixtures.fixture_2hive_8gpu()
+    fix2.devices[4].process_count = 3
+    core.register_node("n0", fix2)
+    res = core.filter({"Pod": _pod(8), "NodeNames": ["n0"]})
+    assert res["NodeNames"] == []
+
+    # refresh with a clean inventory frees everything
+    core.register_node("n0", fixtures.fixture_2hive_8gpu())
+    res = core.filter({"Pod": _pod(8), "NodeNames": ["n0"]})
+    assert res["NodeNames"] == ["n0"]
