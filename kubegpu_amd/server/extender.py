@@ -266,3 +266,26 @@ def serve(
     t.start()
     utils.logf(1, "scheduler extender serving on %s:%d", host, port)
     return server, core
+
+
+def main(argv=None) -> int:
+    import argparse
+    import signal
+
+    ap = argparse.ArgumentParser(prog="kubegpu-amd-extender")
+    ap.add_argument("--host", default="0.0.0.0")
+    ap.add_argument("--port", type=int, default=9109)
+    args = ap.parse_args(argv)
+    server, _core = serve(host=args.host, port=args.port)
+    stop = threading.Event()
+    signal.signal(signal.SIGTERM, lambda *_: stop.set())
+    signal.signal(signal.SIGINT, lambda *_: stop.set())
+    stop.wait()
+    server.shutdown()
+    return 0
+
+
+if __name__ == "__main__":
+    import sys
+
+    sys.exit(main())
