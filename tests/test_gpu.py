@@ -329,3 +329,24 @@ def test_extender_with_real_inventory():
         assert pri[0]["Score"] == 10
     finally:
         server.shutdown()
+
+
+def test_amddevs_cdi_real_device_nodes():
+    """CDI spec from the real inventory: every device node it names
+    exists on the box (the spec is directly consumable by containerd)."""
+    import subprocess as _sp
+    import sys as _sys
+
+    res = _sp.run(
+        [_sys.executable, "-m", "kubegpu_amd.cli.amddevs", "--cdi"],
+        capture_output=True, timeout=120, cwd=REPO, text=True,
+    )
+    assert res.returncode == 0, res.stderr[-1000:]
+    spec = json.loads(res.stdout)
+    assert spec["kind"] == "amd.com/gpu"
+    for n in spec["containerEdits"]["deviceNodes"]:
+        assert os.path.exists(n["path"]), n
+    assert spec["devices"]
+    for dev in spec["devices"]:
+        for n in dev["containerEdits"]["deviceNodes"]:
+            assert os.path.exists(n["path"]), (dev["name"], n)
