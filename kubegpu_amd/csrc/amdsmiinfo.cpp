@@ -21,6 +21,8 @@
 
 #include <amd_smi/amdsmi.h>
 
+#include <sys/stat.h>
+
 #include <cinttypes>
 #include <cstdio>
 #include <cstring>
@@ -153,8 +155,16 @@ bool collect(std::vector<Gpu>& gpus, std::string& driver_version) {
       if (amdsmi_get_gpu_enumeration_info(h, &en) == AMDSMI_STATUS_SUCCESS) {
         if (en.drm_render != 0xFFFFFFFFu && en.drm_render != 0)
           g.render_path = "/dev/dri/renderD" + std::to_string(en.drm_render);
-        if (en.drm_card != 0xFFFFFFFFu)
-          g.card_path = "/dev/dri/card" + std::to_string(en.drm_card);
+        if (en.drm_card != 0xFFFFFFFFu) {
+          // only advertise the card node when it actually exists in
+          // this namespace: containerized nodes often inject renderD*
+          // but not card* — a DeviceSpec naming a missing node fails
+          // container create (caught by the round-2 CDI-on-hardware
+          // test)
+          std::string card = "/dev/dri/card" + std::to_string(en.drm_card);
+          struct stat st {};
+          if (stat(card.c_str(), &st) == 0) g.card_path = card;
+        }
       }
 
       uint64_t vram = 0;

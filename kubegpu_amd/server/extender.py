@@ -81,11 +81,13 @@ class ExtenderCore:
 
         Occupancy comes WITH the registration: the agent reports which
         uuids hold live allocations (its in_use view, reconciled from
-        kubelet's pod-resources API) as a top-level ``in_use`` list,
-        and any GPU with live compute processes counts as occupied too
-        — the extender itself never observes bindings (kube-scheduler
+        kubelet's pod-resources API) as a top-level ``in_use`` list —
+        the extender itself never observes bindings (kube-scheduler
         does not call back after scheduling), so each refresh carries
-        the node's current truth.
+        the node's current truth.  Raw process_count is deliberately
+        NOT treated as occupancy here: system daemons register on KFD
+        (a known-idle box reports process_count 2), so it stays a soft
+        idle-preference signal in GetPreferredAllocation only.
         """
         with self._lock:
             ni = NodeInfo(name=name)
@@ -100,10 +102,9 @@ class ExtenderCore:
                 self.cluster.remove_node(name)
             self.cluster.add_node(ni, mgr._last_info, mgr)
             state = self.cluster.core.nodes[name]
-            occupied = set(in_use)
-            for g in gpus_info.devices:
-                if g.uuid in occupied or (g.process_count or 0) > 0:
-                    state.mark_used(g.uuid)
+            for uuid in set(in_use):
+                if uuid in state.gpus:
+                    state.mark_used(uuid)
             self.cluster.reindex_node(name)
 
     def remove_node(self, name: str) -> None:
