@@ -67,10 +67,15 @@ def add_node_to_sorted_tree_node(parent: SortedTreeNode, node: SortedTreeNode) -
 
 
 def compare_tree_node(a: Optional[SortedTreeNode], b: Optional[SortedTreeNode]) -> bool:
-    """Structural equality (val, score, recursively ordered children)."""
+    """Structural equality: val + recursively ordered children.
+
+    Scores are deliberately NOT compared — exactly the reference
+    (CompareTreeNode, typeutils.go:75-93, checks Val and Child only):
+    score is derived from the shape, so two identically-shaped trees
+    are the same canonical tree even mid-rescoring."""
     if a is None or b is None:
         return a is b
-    if a.val != b.val or a.score != b.score:
+    if a.val != b.val:
         return False
     if len(a.children) != len(b.children):
         return False

@@ -60,3 +60,14 @@ def test_print_tree_node_renders():
     add_to_sorted_tree_node(root, 2)
     out = print_tree_node(root)
     assert "val=2" in out
+
+
+def test_compare_ignores_score_like_reference():
+    """CompareTreeNode (typeutils.go:75-93) checks Val and Child only —
+    score is derived, so identically-shaped trees compare equal even
+    with different scores."""
+    a = SortedTreeNode(val=4, score=1.0)
+    b = SortedTreeNode(val=4, score=99.0)
+    add_to_sorted_tree_node(a, 4)
+    add_to_sorted_tree_node(b, 4)
+    assert compare_tree_node(a, b)
