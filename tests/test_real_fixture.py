@@ -85,3 +85,32 @@ def test_round2_capture_parses_with_runtime_versions():
     ni = NodeInfo(name="captured-r2")
     mgr.update_node_info(ni)
     assert ni.kube_alloc[RESOURCE_GPU] == 1
+
+
+def test_post_cardfix_capture_omits_phantom_card_node():
+    """Verbatim capture from the card-fix binary (session P): the
+    containerized box injects renderD but not card*, so card_path must
+    be empty — the manager and CDI then never name a missing node."""
+    path = os.path.join(os.path.dirname(__file__), "fixtures",
+                        "real_1x_mi355x_r2b.json")
+    with open(path) as f:
+        info = GpusInfo.from_json(f.read())
+    g = info.devices[0]
+    assert g.card_path == ""
+    assert g.render_path.startswith("/dev/dri/renderD")
+    # allocate path: /dev/kfd + render only, no phantom card
+    mgr = create_device_plugin(FakeBackend(info))
+    cluster = Cluster()
+    cluster.add_node_from_manager("n", mgr)
+    pod = PodInfo(name="p", running_containers={
+        "c": ContainerInfo(kube_requests={RESOURCE_GPU: 1})})
+    cluster.schedule(pod)
+    _, devices, _ = cluster.container_allocate(pod, "c")
+    assert devices[0] == "/dev/kfd"
+    assert all("card" not in d for d in devices[1:])
+    # CDI spec from the same capture names no card node either
+    from kubegpu_amd.deviceplugin.cdi import cdi_spec
+    spec = cdi_spec(info)
+    for dev in spec["devices"]:
+        for n in dev["containerEdits"]["deviceNodes"]:
+            assert "card" not in n["path"]
